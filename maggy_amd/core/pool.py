@@ -1,0 +1,155 @@
+"""TrialPool: process-per-GPU worker management.
+
+Replaces the reference's Spark executor fan-out + Reservations table
+(/root/reference/maggy/core/rpc.py:45-123, spark_driver.py:136-145).
+One worker process per GPU (pinned via HIP_VISIBLE_DEVICES), a Pipe per
+worker for control messages and a shared-memory MetricRing per worker for
+the metric stream and the mid-trial stop word.
+
+Lost-worker semantics (parity with the BLACK message, rpc.py:415-437):
+``reap()`` detects dead worker processes, and the driver respawns the
+worker and re-queues the trial it held.
+"""
+import multiprocessing as mp
+from multiprocessing.connection import wait as conn_wait
+
+from maggy_amd import constants
+from maggy_amd.core import messages as M
+from maggy_amd.core.shm import MetricRing
+from maggy_amd.core.worker import worker_main
+
+
+class WorkerHandle:
+    def __init__(self, worker_id, gpu_id):
+        self.worker_id = worker_id
+        self.gpu_id = gpu_id
+        self.process = None
+        self.conn = None  # driver end of the pipe
+        self.ring = None
+        self.trial_id = None  # trial currently assigned (driver view)
+        self.registered = False
+        self.respawns = 0
+
+
+class TrialPool:
+    def __init__(self, num_workers, log_dir, payload, gpu_ids=None,
+                 ring_slots=constants.SCHEDULER.RING_SLOTS, start_method="spawn"):
+        self.num_workers = num_workers
+        self.log_dir = log_dir
+        self.payload = payload
+        self.ring_slots = ring_slots
+        self.ctx = mp.get_context(start_method)
+        if gpu_ids is None:
+            gpu_ids = [None] * num_workers
+        self.workers = [WorkerHandle(i, gpu_ids[i]) for i in range(num_workers)]
+
+    # -- lifecycle ------------------------------------------------------
+    def start(self):
+        for w in self.workers:
+            self._spawn(w)
+
+    def _spawn(self, w):
+        w.ring = MetricRing(slots=self.ring_slots, create=True)
+        parent_conn, child_conn = self.ctx.Pipe()
+        w.conn = parent_conn
+        w.registered = False
+        w.process = self.ctx.Process(
+            target=worker_main,
+            args=(w.worker_id, w.gpu_id, child_conn, w.ring.name,
+                  self.ring_slots, self.log_dir, self.payload),
+            daemon=True,
+        )
+        w.process.start()
+        child_conn.close()
+
+    def respawn(self, w):
+        """Respawn a dead worker, preserving its id/GPU (BLACK semantics)."""
+        try:
+            if w.ring is not None:
+                w.ring.close()
+                w.ring.unlink()
+        except Exception:
+            pass
+        w.respawns += 1
+        w.trial_id = None
+        self._spawn(w)
+
+    def shutdown(self, timeout=constants.SCHEDULER.JOIN_TIMEOUT):
+        for w in self.workers:
+            try:
+                if w.process is not None and w.process.is_alive():
+                    w.conn.send((M.GSTOP,))
+            except (BrokenPipeError, OSError):
+                pass
+        for w in self.workers:
+            if w.process is not None:
+                w.process.join(timeout=timeout)
+                if w.process.is_alive():
+                    w.process.terminate()
+                    w.process.join(timeout=5)
+        for w in self.workers:
+            try:
+                if w.ring is not None:
+                    w.ring.close()
+                    w.ring.unlink()
+            except Exception:
+                pass
+            try:
+                if w.conn is not None:
+                    w.conn.close()
+            except Exception:
+                pass
+
+    # -- event-loop helpers ---------------------------------------------
+    def poll_messages(self, timeout=constants.SCHEDULER.POLL_TIMEOUT):
+        """Block up to ``timeout`` for control messages; return list of
+        (worker, message) pairs. Dead pipes are skipped (reap() handles)."""
+        conns = {w.conn: w for w in self.workers
+                 if w.conn is not None and w.process is not None}
+        if not conns:
+            return []
+        ready = conn_wait(list(conns.keys()), timeout=timeout)
+        out = []
+        for c in ready:
+            w = conns[c]
+            try:
+                while c.poll():
+                    out.append((w, c.recv()))
+            except (EOFError, OSError):
+                pass  # worker died; reap() will notice
+        return out
+
+    def drain_metrics(self):
+        """Drain every worker's metric ring; return list of
+        (worker, tag, step, value)."""
+        out = []
+        for w in self.workers:
+            if w.ring is None:
+                continue
+            for tag, step, value in w.ring.drain():
+                out.append((w, tag, step, value))
+        return out
+
+    def reap(self):
+        """Return workers whose process died (exitcode set)."""
+        dead = []
+        for w in self.workers:
+            if w.process is not None and not w.process.is_alive():
+                dead.append(w)
+        return dead
+
+    # -- assignment ------------------------------------------------------
+    def assign(self, w, trial):
+        w.trial_id = trial.trial_id
+        w.ring.clear_stop()
+        w.conn.send((M.TRIAL, trial.trial_id, trial.params))
+
+    def request_stop(self, trial_id):
+        """Flag a running trial for early stop via its worker's stop word."""
+        from maggy_amd.core.shm import trial_tag
+
+        for w in self.workers:
+            if w.trial_id == trial_id and w.ring is not None:
+                w.ring.set_stop(trial_tag(trial_id))
+                return True
+        return False

@@ -1,0 +1,156 @@
+"""Trial-pool worker process: runs user training functions on one GPU.
+
+Replaces the reference's Spark-task executor wrapper
+(/root/reference/maggy/core/executors/trial_executor.py:35-213) with a
+long-lived process pinned to a single MI355X via ``HIP_VISIBLE_DEVICES``.
+Control messages (TRIAL / FINAL / ERROR / GSTOP) travel over a
+multiprocessing Pipe; the hot metric stream and mid-trial STOP go through
+the lock-free shared-memory ring (core/shm.py).
+
+The per-trial semantics mirror the reference executor exactly:
+  - make/clean the trial dir, write ``.hparams.json``
+  - tee ``print`` into the reporter log
+  - call ``train_fn`` with signature-inspected kwargs
+    (model/dataset/hparams/reporter + ablation generators)
+  - normalize + persist the return value (``.outputs.json``/``.metric``)
+  - catch ``EarlyStopException`` and finalize with the last metric
+"""
+import builtins
+import json
+import os
+import time
+import traceback
+
+from maggy_amd import util
+from maggy_amd.core import messages as M
+from maggy_amd.core.reporter import Reporter
+from maggy_amd.core.shm import MetricRing
+from maggy_amd.exceptions import EarlyStopException
+from maggy_amd.utils.jsonutil import json_default_numpy
+
+
+def _pin_gpu(gpu_id):
+    if gpu_id is None:
+        return
+    os.environ["HIP_VISIBLE_DEVICES"] = str(gpu_id)
+    os.environ["CUDA_VISIBLE_DEVICES"] = str(gpu_id)
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
+
+def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload):
+    """Entry point of a spawned trial worker.
+
+    ``payload`` is a dict pickled once at pool start (never per trial):
+    {train_fn, model, dataset, optimization_key, experiment_type,
+     dataset_generator, model_generator}.
+    """
+    _pin_gpu(gpu_id)
+    ring = MetricRing(name=ring_name, slots=ring_slots)
+    log_file = os.path.join(log_dir, "executor_{}.log".format(worker_id))
+    real_print = builtins.print
+    reporter = Reporter(
+        ring=ring, log_file=log_file, worker_id=worker_id, print_fn=real_print
+    )
+
+    def maggy_print(*args, **kwargs):
+        real_print(*args, **kwargs)
+        reporter.log(" ".join(str(x) for x in args), True)
+
+    train_fn = payload["train_fn"]
+    optimization_key = payload.get("optimization_key", "Metric")
+    experiment_type = payload.get("experiment_type", "optimization")
+
+    try:
+        conn.send((M.REG, worker_id, os.getpid()))
+        while True:
+            msg = conn.recv()
+            if msg[0] == M.GSTOP:
+                break
+            if msg[0] != M.TRIAL:
+                continue
+            _, trial_id, parameters = msg
+            parameters = dict(parameters)
+            ablation_params = None
+            if experiment_type == "ablation":
+                ablation_params = {
+                    "ablated_feature": parameters.pop("ablated_feature", "None"),
+                    "ablated_layer": parameters.pop("ablated_layer", "None"),
+                }
+
+            trial_dir = os.path.join(log_dir, trial_id)
+            trial_log_file = os.path.join(trial_dir, "output.log")
+            reporter.set_trial_id(trial_id)
+            if os.path.exists(trial_dir):
+                util.clean_dir(trial_dir, keep=[trial_log_file])
+            else:
+                os.makedirs(trial_dir, exist_ok=True)
+            reporter.init_logger(trial_log_file)
+            with open(os.path.join(trial_dir, ".hparams.json"), "w") as f:
+                f.write(
+                    json.dumps(
+                        ablation_params if ablation_params else parameters,
+                        default=json_default_numpy,
+                    )
+                )
+
+            start = time.time()
+            early_stopped = False
+            builtins.print = maggy_print
+            try:
+                reporter.log("Starting Trial: {}".format(trial_id), False)
+                reporter.log("Trial Configuration: {}".format(parameters), False)
+                extra = {}
+                if experiment_type == "ablation":
+                    gen_extra = _ablation_generators(payload, parameters,
+                                                     ablation_params)
+                    extra.update(gen_extra)
+                kwargs = util.build_train_kwargs(
+                    train_fn,
+                    model=extra.pop("model", payload.get("model")),
+                    dataset=extra.pop("dataset", payload.get("dataset")),
+                    hparams=parameters,
+                    reporter=reporter,
+                    extra=extra,
+                )
+                retval = train_fn(**kwargs)
+                retval = util.handle_return_val(
+                    retval, trial_dir, optimization_key, trial_log_file
+                )
+                conn.send((M.FINAL, worker_id, trial_id, retval,
+                           time.time() - start, early_stopped, reporter.logs))
+            except EarlyStopException as e:
+                early_stopped = True
+                reporter.log("Early Stopped Trial.", False)
+                conn.send((M.FINAL, worker_id, trial_id, e.metric,
+                           time.time() - start, early_stopped, reporter.logs))
+            except Exception:
+                tb = traceback.format_exc()
+                reporter.log(tb, False)
+                conn.send((M.ERROR, worker_id, trial_id, tb))
+            finally:
+                builtins.print = real_print
+                reporter.logs = ""
+                reporter.reset()
+    finally:
+        builtins.print = real_print
+        reporter.close_logger()
+        ring.close()
+        conn.close()
+
+
+def _ablation_generators(payload, parameters, ablation_params):
+    """Resolve the per-trial model/dataset for an ablation trial via the
+    generators shipped in the payload (parity: loco.py generator contract,
+    realized for PyTorch nn.Modules instead of Keras JSON surgery)."""
+    extra = {}
+    model_generator = payload.get("model_generator")
+    dataset_generator = payload.get("dataset_generator")
+    if model_generator is not None:
+        extra["model"] = model_generator(
+            ablated_layer=ablation_params.get("ablated_layer", "None")
+        )
+    if dataset_generator is not None:
+        extra["dataset"] = dataset_generator(
+            ablated_feature=ablation_params.get("ablated_feature", "None")
+        )
+    return extra

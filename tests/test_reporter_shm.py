@@ -1,0 +1,95 @@
+import pytest
+
+from maggy_amd.core.reporter import Reporter
+from maggy_amd.core.shm import MetricRing, trial_tag
+from maggy_amd.exceptions import (
+    BroadcastMetricTypeError,
+    BroadcastStepValueError,
+    EarlyStopException,
+)
+
+
+def make_ring():
+    return MetricRing(slots=64, create=True)
+
+
+def test_ring_push_drain():
+    ring = make_ring()
+    try:
+        for i in range(10):
+            ring.push(7, i, float(i) * 0.5)
+        recs = ring.drain()
+        assert recs == [(7, i, i * 0.5) for i in range(10)]
+        assert ring.drain() == []
+        ring.push(7, 10, 5.0)
+        assert ring.drain() == [(7, 10, 5.0)]
+    finally:
+        ring.close()
+        ring.unlink()
+
+
+def test_ring_overrun_keeps_newest():
+    ring = MetricRing(slots=8, create=True)
+    try:
+        for i in range(20):
+            ring.push(1, i, float(i))
+        recs = ring.drain()
+        assert len(recs) == 8
+        assert recs[-1] == (1, 19, 19.0)
+        assert recs[0] == (1, 12, 12.0)
+    finally:
+        ring.close()
+        ring.unlink()
+
+
+def test_broadcast_validation_and_ring():
+    ring = make_ring()
+    try:
+        rep = Reporter(ring=ring)
+        rep.set_trial_id("3d1cc9fdb1d4d001")
+        rep.broadcast(1.0)          # step auto-increments to 0
+        rep.broadcast(2.0, 5)
+        with pytest.raises(BroadcastMetricTypeError):
+            rep.broadcast("nope")
+        with pytest.raises(BroadcastStepValueError):
+            rep.broadcast(3.0, 2)   # non-monotone
+        tag = trial_tag("3d1cc9fdb1d4d001")
+        assert ring.drain() == [(tag, 0, 1.0), (tag, 5, 2.0)]
+    finally:
+        ring.close()
+        ring.unlink()
+
+
+def test_stop_word_raises_early_stop_only_for_current_trial():
+    ring = make_ring()
+    try:
+        rep = Reporter(ring=ring)
+        rep.set_trial_id("3d1cc9fdb1d4d001")
+        rep.broadcast(1.0, 0)
+        # stop aimed at a DIFFERENT trial: no effect
+        ring.set_stop(trial_tag("aaaaaaaaaaaaaaaa"))
+        rep.broadcast(2.0, 1)
+        # stop aimed at this trial: EarlyStopException on next broadcast
+        ring.set_stop(trial_tag("3d1cc9fdb1d4d001"))
+        with pytest.raises(EarlyStopException) as ei:
+            rep.broadcast(3.0, 2)
+        assert ei.value.metric == 3.0
+    finally:
+        ring.close()
+        ring.unlink()
+
+
+def test_local_early_stop_requires_metric():
+    rep = Reporter()
+    rep.early_stop()          # no metric yet -> ignored
+    rep.broadcast(1.0, 0)
+    rep.early_stop()
+    with pytest.raises(EarlyStopException):
+        rep.broadcast(2.0, 1)
+
+
+def test_reset():
+    rep = Reporter()
+    rep.broadcast(1.0, 0)
+    rep.reset()
+    assert rep.metric is None and rep.step == -1 and not rep.stop
