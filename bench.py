@@ -1,0 +1,223 @@
+#!/usr/bin/env python3
+"""maggy_amd flagship benchmark (driver contract).
+
+Default: ResNet-50 synthetic 224x224 bf16 training step (the BASELINE.json
+headline: samples/sec/GPU at 1/2/4/8 GPUs, weak scaling) — forward +
+backward + fused HIP Adam step, DDP over RCCL/xGMI for N>1.
+
+    python bench.py --gpus N --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Other modes:
+    --mode asha     completed ASHA trials/hr on the trial pool (1 proc)
+    --model llama8b Llama-3-8B bf16 DP step (tokens/sec)
+
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=0,
+                   help="per-GPU batch (0 = model default)")
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "llama8b", "llama1b"])
+    p.add_argument("--mode", default="train", choices=["train", "asha"])
+    p.add_argument("--seq-len", type=int, default=4096)
+    return p.parse_args()
+
+
+def dist_env():
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    return rank, world, local_rank
+
+
+def setup_dist(world, local_rank):
+    import torch.distributed as dist
+
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+    return dist
+
+
+def make_resnet_step(args, device, world):
+    from maggy_amd.models import resnet50
+    from maggy_amd.ops import FusedAdam
+
+    batch = args.batch or 256
+    model = resnet50().to(device, memory_format=torch.channels_last)
+    if world > 1:
+        from maggy_amd.parallel.dist import wrap_ddp
+
+        model = wrap_ddp(model)
+    opt = FusedAdam(model.parameters(), lr=1e-3)
+    x = torch.randn(batch, 3, 224, 224, device=device).to(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (batch,), device=device)
+    loss_fn = torch.nn.CrossEntropyLoss()
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            loss = loss_fn(model(x), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    cfg = {"model": "resnet50", "global_batch": batch * world,
+           "image_size": 224, "parallelism": "dp{}".format(world)}
+    return step, batch, "samples/sec (ResNet-50 synthetic 224x224)", cfg
+
+
+def make_llama_step(args, device, world, size):
+    from maggy_amd.models import LlamaConfig, LlamaModel
+    from maggy_amd.ops import FusedAdam
+
+    cfg_model = (LlamaConfig.llama3_8b() if size == "8b"
+                 else LlamaConfig.small_1b())
+    batch = args.batch or (2 if size == "8b" else 8)
+    seq = args.seq_len
+    # construct directly on the GPU (init kernels run on-device; 8B fp32
+    # transient fits easily in 288 GB HBM3E), then cast params to bf16
+    with torch.device(device):
+        model = LlamaModel(cfg_model)
+    model = model.to(torch.bfloat16)
+    model.rope_cos = model.rope_cos.float()
+    model.rope_sin = model.rope_sin.float()
+    if world > 1:
+        from maggy_amd.parallel.dist import wrap_ddp
+
+        model = wrap_ddp(model)
+    opt = FusedAdam(model.parameters(), lr=1e-4, max_grad_norm=1.0)
+    tokens = torch.randint(0, cfg_model.vocab_size, (batch, seq),
+                           device=device)
+    targets = torch.randint(0, cfg_model.vocab_size, (batch, seq),
+                            device=device)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        loss = model(tokens, targets)
+        loss.backward()
+        opt.step()
+        return loss
+
+    cfg = {"model": "llama3-{}".format(size), "global_batch": batch * world,
+           "seq_len": seq, "parallelism": "dp{}".format(world)}
+    return step, batch * seq, "tokens/sec (Llama-3 bf16 synthetic)", cfg
+
+
+def run_train(args):
+    rank, world, local_rank = dist_env()
+    dist = setup_dist(world, local_rank)
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+    torch.backends.cudnn.benchmark = True
+
+    if args.model == "resnet50":
+        step, per_step_items, metric, cfg = make_resnet_step(
+            args, device, world)
+    else:
+        size = "8b" if args.model == "llama8b" else "1b"
+        step, per_step_items, metric, cfg = make_llama_step(
+            args, device, world, size)
+
+    for _ in range(args.warmup):
+        step()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    value = per_step_items * world * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    if rank == 0:
+        print(json.dumps({
+            "metric": metric,
+            "value": round(value, 2),
+            "unit": metric.split(" ")[0],
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": cfg,
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+def run_asha(args):
+    """Completed ASHA trials/hr on the trial pool (reference headline:
+    async trial scheduling throughput)."""
+    from maggy_amd import Searchspace, experiment
+    from maggy_amd.config import HyperparameterOptConfig
+    from bench_trials import resnet_trial_fn
+
+    n_gpus = min(args.gpus, torch.cuda.device_count())
+    sp = Searchspace(lr=("DOUBLE", [1e-4, 1e-2]),
+                     momentum=("DOUBLE", [0.8, 0.99]))
+    num_trials = max(16, 4 * n_gpus)
+    cfg = HyperparameterOptConfig(
+        num_trials=num_trials, optimizer="asha", searchspace=sp,
+        direction="min", es_policy="median", es_min=4,
+        num_workers=n_gpus, name="bench-asha")
+    t0 = time.time()
+    res = experiment.lagom(resnet_trial_fn, cfg)
+    elapsed = time.time() - t0
+    trials_per_hr = res["num_trials"] / elapsed * 3600.0
+    print(json.dumps({
+        "metric": "completed trials/hr (ASHA, ResNet-50 synthetic)",
+        "value": round(trials_per_hr, 2),
+        "unit": "trials/hr",
+        "n_gpus": n_gpus,
+        "steps": res["num_trials"],
+        "warmup": 0,
+        "ms_per_step": round(elapsed / res["num_trials"] * 1000.0, 1),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "config": {"model": "resnet50", "num_trials": res["num_trials"],
+                   "optimizer": "asha", "parallelism":
+                   "trialpool{}".format(n_gpus)},
+    }))
+
+
+if __name__ == "__main__":
+    args = parse_args()
+    if args.mode == "asha":
+        run_asha(args)
+    else:
+        run_train(args)

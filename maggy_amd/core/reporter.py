@@ -65,6 +65,7 @@ class Reporter:
         for early stop (checked on every call; delivery latency is one
         driver event-loop iteration).
         """
+        metric = self._to_scalar(metric)
         with self.lock:
             if step is None:
                 step = self.step + 1
@@ -89,6 +90,19 @@ class Reporter:
                 stopped = True
             if stopped:
                 raise exceptions.EarlyStopException(metric)
+
+    @staticmethod
+    def _to_scalar(metric):
+        """Allow broadcast() to take a torch tensor: a 0-d tensor reads its
+        item; a larger CUDA tensor is mean-reduced by the HIP reduction
+        kernel (reference call-site N8)."""
+        if type(metric).__module__.startswith("torch"):
+            if metric.numel() == 1:
+                return float(metric.detach().item())
+            from maggy_amd.ops import metric_mean
+
+            return metric_mean(metric.detach())
+        return metric
 
     def log(self, log_msg, jupyter=False):
         """Log to the worker logfile (and the trial logfile when a trial is

@@ -1,0 +1,167 @@
+"""Llama-3-style decoder-only transformer (RMSNorm / SwiGLU / RoPE / GQA).
+
+BASELINE.json config 4: Llama-3 8B bf16 data-parallel over 8x MI355X.
+Self-contained (no transformers dependency); random-init weights, synthetic
+token data.  Attention goes through torch.nn.functional
+scaled_dot_product_attention, which on ROCm dispatches to the fused
+(AOTriton/composable-kernel) kernels when available and the math path
+otherwise.  The 288 GB HBM3E per GPU fits the full 8B model + Adam states
+without sharding, so plain DP is the natural parallelism.
+"""
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    dim: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    ffn_hidden: int = 14336
+    max_seq_len: int = 8192
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+
+    @staticmethod
+    def llama3_8b():
+        return LlamaConfig()
+
+    @staticmethod
+    def tiny(vocab_size=512):
+        """CPU-testable toy config."""
+        return LlamaConfig(vocab_size=vocab_size, dim=64, n_layers=2,
+                           n_heads=4, n_kv_heads=2, ffn_hidden=128,
+                           max_seq_len=128)
+
+    @staticmethod
+    def small_1b():
+        """Llama-3.2-1B-ish shape for single-GPU experiments."""
+        return LlamaConfig(vocab_size=128256, dim=2048, n_layers=16,
+                           n_heads=32, n_kv_heads=8, ffn_hidden=8192)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim, eps=1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        dt = x.dtype
+        x = x.float()
+        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (x * self.weight.float()).to(dt)
+
+
+def precompute_rope(dim, max_seq_len, theta):
+    inv_freq = 1.0 / (theta ** (torch.arange(0, dim, 2).float() / dim))
+    t = torch.arange(max_seq_len).float()
+    freqs = torch.outer(t, inv_freq)
+    return torch.cos(freqs), torch.sin(freqs)
+
+
+def apply_rope(x, cos, sin):
+    # x: [B, H, T, D]; rotate pairs (x0,x1) in the last dim
+    T = x.shape[-2]
+    cos = cos[:T].to(x.dtype)
+    sin = sin[:T].to(x.dtype)
+    x1, x2 = x[..., 0::2], x[..., 1::2]
+    out = torch.empty_like(x)
+    out[..., 0::2] = x1 * cos - x2 * sin
+    out[..., 1::2] = x2 * cos + x1 * sin
+    return out
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.n_heads = cfg.n_heads
+        self.n_kv_heads = cfg.n_kv_heads
+        self.head_dim = cfg.dim // cfg.n_heads
+        self.wq = nn.Linear(cfg.dim, cfg.n_heads * self.head_dim, bias=False)
+        self.wk = nn.Linear(cfg.dim, cfg.n_kv_heads * self.head_dim,
+                            bias=False)
+        self.wv = nn.Linear(cfg.dim, cfg.n_kv_heads * self.head_dim,
+                            bias=False)
+        self.wo = nn.Linear(cfg.n_heads * self.head_dim, cfg.dim, bias=False)
+
+    def forward(self, x, cos, sin):
+        B, T, _ = x.shape
+        q = self.wq(x).view(B, T, self.n_heads, self.head_dim).transpose(1, 2)
+        k = self.wk(x).view(B, T, self.n_kv_heads, self.head_dim).transpose(1, 2)
+        v = self.wv(x).view(B, T, self.n_kv_heads, self.head_dim).transpose(1, 2)
+        q = apply_rope(q, cos, sin)
+        k = apply_rope(k, cos, sin)
+        out = F.scaled_dot_product_attention(
+            q, k, v, is_causal=True, enable_gqa=(self.n_kv_heads != self.n_heads)
+        )
+        out = out.transpose(1, 2).reshape(B, T, -1)
+        return self.wo(out)
+
+
+class FeedForward(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.w1 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False)  # gate
+        self.w3 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False)  # up
+        self.w2 = nn.Linear(cfg.ffn_hidden, cfg.dim, bias=False)  # down
+
+    def forward(self, x):
+        return self.w2(F.silu(self.w1(x)) * self.w3(x))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.attn = Attention(cfg)
+        self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.ffn = FeedForward(cfg)
+
+    def forward(self, x, cos, sin):
+        x = x + self.attn(self.attn_norm(x), cos, sin)
+        x = x + self.ffn(self.ffn_norm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.cfg = cfg
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.layers = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        cos, sin = precompute_rope(cfg.dim // cfg.n_heads, cfg.max_seq_len,
+                                   cfg.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.apply(self._init)
+
+    def _init(self, m):
+        if isinstance(m, nn.Linear):
+            nn.init.normal_(m.weight, std=0.02 / math.sqrt(
+                2 * self.cfg.n_layers))
+        elif isinstance(m, nn.Embedding):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, tokens, targets=None):
+        x = self.tok_emb(tokens)
+        for layer in self.layers:
+            x = layer(x, self.rope_cos, self.rope_sin)
+        x = self.norm(x)
+        if targets is not None:
+            logits = self.lm_head(x)
+            return F.cross_entropy(
+                logits.float().view(-1, logits.size(-1)), targets.view(-1)
+            )
+        return self.lm_head(x)
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())

@@ -1,0 +1,119 @@
+// pybind11 bindings for the maggy_amd HIP kernels (MI355X / gfx950).
+//
+// The chunk/tensor metadata tables are packed in Python (ops/fused_adam.py)
+// and live in device memory as uint8 tensors; these bindings only validate,
+// fetch the current stream and call the launch wrappers defined in
+// maggy_kernels.hip.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+extern "C" void launch_multi_l2norm_sq(const void*, int, const void*, int,
+                                       float*, hipStream_t);
+extern "C" void launch_multi_fused_adam(const void*, int, const void*, int,
+                                        float, float, float, float, float,
+                                        float, float, const float*, float,
+                                        float, hipStream_t);
+extern "C" void launch_multi_fused_sgd(const void*, int, const void*, int,
+                                       float, float, float, float, int, int,
+                                       const float*, float, float,
+                                       hipStream_t);
+extern "C" void launch_reduce_sum_f32(const float*, long long, float*,
+                                      hipStream_t);
+extern "C" void launch_reduce_sum_bf16(const void*, long long, float*,
+                                       hipStream_t);
+extern "C" void launch_reduce_max_f32(const float*, long long, float*,
+                                      hipStream_t);
+
+namespace {
+
+hipStream_t current_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_table(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a CUDA tensor");
+  TORCH_CHECK(t.dtype() == torch::kUInt8, name, " must be uint8");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void multi_l2norm_sq(torch::Tensor chunks, int64_t n_chunks,
+                     torch::Tensor tensors, bool grad_bf16,
+                     torch::Tensor norm_sq) {
+  check_table(chunks, "chunks");
+  check_table(tensors, "tensors");
+  TORCH_CHECK(norm_sq.is_cuda() && norm_sq.dtype() == torch::kFloat32);
+  launch_multi_l2norm_sq(chunks.data_ptr(), (int)n_chunks,
+                         tensors.data_ptr(), grad_bf16 ? 1 : 0,
+                         norm_sq.data_ptr<float>(), current_stream());
+}
+
+void multi_fused_adam(torch::Tensor chunks, int64_t n_chunks,
+                      torch::Tensor tensors, bool grad_bf16, double lr,
+                      double beta1, double beta2, double eps,
+                      double weight_decay, double bc1, double bc2,
+                      c10::optional<torch::Tensor> norm_sq, double max_norm,
+                      double grad_scale_inv) {
+  check_table(chunks, "chunks");
+  check_table(tensors, "tensors");
+  const float* ns = nullptr;
+  if (norm_sq.has_value()) ns = norm_sq->data_ptr<float>();
+  launch_multi_fused_adam(chunks.data_ptr(), (int)n_chunks,
+                          tensors.data_ptr(), grad_bf16 ? 1 : 0, (float)lr,
+                          (float)beta1, (float)beta2, (float)eps,
+                          (float)weight_decay, (float)bc1, (float)bc2, ns,
+                          (float)max_norm, (float)grad_scale_inv,
+                          current_stream());
+}
+
+void multi_fused_sgd(torch::Tensor chunks, int64_t n_chunks,
+                     torch::Tensor tensors, bool grad_bf16, double lr,
+                     double momentum, double weight_decay, double dampening,
+                     bool nesterov, bool first_step,
+                     c10::optional<torch::Tensor> norm_sq, double max_norm,
+                     double grad_scale_inv) {
+  check_table(chunks, "chunks");
+  check_table(tensors, "tensors");
+  const float* ns = nullptr;
+  if (norm_sq.has_value()) ns = norm_sq->data_ptr<float>();
+  launch_multi_fused_sgd(chunks.data_ptr(), (int)n_chunks,
+                         tensors.data_ptr(), grad_bf16 ? 1 : 0, (float)lr,
+                         (float)momentum, (float)weight_decay,
+                         (float)dampening, nesterov ? 1 : 0,
+                         first_step ? 1 : 0, ns, (float)max_norm,
+                         (float)grad_scale_inv, current_stream());
+}
+
+void reduce_sum(torch::Tensor in, torch::Tensor out) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous());
+  TORCH_CHECK(out.is_cuda() && out.dtype() == torch::kFloat32);
+  if (in.dtype() == torch::kFloat32) {
+    launch_reduce_sum_f32(in.data_ptr<float>(), in.numel(),
+                          out.data_ptr<float>(), current_stream());
+  } else if (in.dtype() == torch::kBFloat16) {
+    launch_reduce_sum_bf16(in.data_ptr(), in.numel(),
+                           out.data_ptr<float>(), current_stream());
+  } else {
+    TORCH_CHECK(false, "reduce_sum: dtype must be float32 or bfloat16");
+  }
+}
+
+void reduce_max(torch::Tensor in, torch::Tensor out) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous());
+  TORCH_CHECK(in.dtype() == torch::kFloat32,
+              "reduce_max: float32 only");
+  launch_reduce_max_f32(in.data_ptr<float>(), in.numel(),
+                        out.data_ptr<float>(), current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("multi_l2norm_sq", &multi_l2norm_sq,
+        "sum of squares over the chunk table -> norm_sq[0]");
+  m.def("multi_fused_adam", &multi_fused_adam,
+        "fused multi-tensor AdamW step with in-kernel grad clip");
+  m.def("multi_fused_sgd", &multi_fused_sgd,
+        "fused multi-tensor SGD(+momentum) step with in-kernel grad clip");
+  m.def("reduce_sum", &reduce_sum, "scalar sum reduction");
+  m.def("reduce_max", &reduce_max, "scalar max reduction");
+}

@@ -1,0 +1,65 @@
+"""Distributed data-parallel plumbing: RCCL over xGMI.
+
+Replaces the reference's torch_dist_executor bring-up
+(/root/reference/maggy/core/executors/torch_dist_executor.py:247-285 and
+patching/modules.py:38-65): process-per-GPU ranks over torch.distributed
+("nccl" backend IS RCCL on ROCm), DDP with xGMI-sized gradient buckets.
+
+xGMI topology notes: each MI355X has 7 point-to-point
+links (~153 GB/s each) to the other 7 GPUs — fully connected, no switch.
+A ring all-reduce is per-link bound, so fewer/larger buckets win: bucket
+transfer time must dominate launch+sync overhead.  Default bucket is 64 MB
+(vs DDP's 25 MB) with gradient_as_bucket_view to avoid the copy; measured
+tuning happens in bench profiles.
+"""
+import datetime
+import os
+
+import torch
+import torch.distributed as dist
+
+XGMI_BUCKET_CAP_MB = 64
+
+
+def init_process_group(backend=None, timeout_s=120):
+    """Initialize torch.distributed from the standard env vars
+    (RANK/WORLD_SIZE/LOCAL_RANK/MASTER_ADDR/MASTER_PORT)."""
+    if dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    dist.init_process_group(
+        backend, timeout=datetime.timedelta(seconds=timeout_s))
+    rank = dist.get_rank()
+    if backend == "nccl":
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    return rank, dist.get_world_size()
+
+
+def wrap_ddp(module, bucket_cap_mb=None, device_ids=None,
+             find_unused_parameters=False):
+    """DDP wrap with xGMI-tuned bucketing (bucketed all-reduce overlapped
+    with backward — reference call-site N2)."""
+    if bucket_cap_mb is None:
+        bucket_cap_mb = XGMI_BUCKET_CAP_MB
+    kwargs = dict(
+        bucket_cap_mb=bucket_cap_mb,
+        gradient_as_bucket_view=True,
+        find_unused_parameters=find_unused_parameters,
+    )
+    if torch.cuda.is_available() and next(
+            module.parameters()).is_cuda:
+        kwargs["device_ids"] = device_ids or [torch.cuda.current_device()]
+    return torch.nn.parallel.DistributedDataParallel(module, **kwargs)
+
+
+def barrier():
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def cleanup():
+    if dist.is_initialized():
+        dist.destroy_process_group()
