@@ -1,0 +1,79 @@
+"""Distribution-transparent DataLoader.
+
+Parity: /root/reference/maggy/core/patching/dataloader.py:33-163
+(MaggyDataLoader) — subclasses torch DataLoader, force-injects a
+DistributedSampler when a process group is up, and moves every batch to the
+worker's GPU.  MI355X-native difference (call-site N9): batches are staged
+through pinned memory and copied H2D with non_blocking=True on a dedicated
+HIP copy stream, overlapping the copy with compute.
+"""
+import torch
+from torch.utils.data import DataLoader as TorchDataLoader
+from torch.utils.data.distributed import DistributedSampler
+
+
+class MaggyDataLoader(TorchDataLoader):
+    def __init__(self, dataset, batch_size=1, shuffle=False, **kwargs):
+        import torch.distributed as dist
+
+        sampler = None
+        if dist.is_available() and dist.is_initialized() and \
+                dist.get_world_size() > 1:
+            sampler = DistributedSampler(dataset, shuffle=shuffle)
+            shuffle = False
+        kwargs.pop("sampler", None)
+        pin = torch.cuda.is_available()
+        kwargs.setdefault("pin_memory", pin)
+        super().__init__(dataset, batch_size=batch_size, shuffle=shuffle,
+                         sampler=sampler, **kwargs)
+        self._device = (torch.device("cuda", torch.cuda.current_device())
+                        if torch.cuda.is_available() else None)
+        self._copy_stream = (torch.cuda.Stream()
+                             if self._device is not None else None)
+
+    def __iter__(self):
+        base = super().__iter__()
+        if self._device is None:
+            return base
+        return _DeviceIter(base, self._device, self._copy_stream)
+
+
+class _DeviceIter:
+    """Prefetching H2D iterator: the next batch's copy runs on a separate
+    stream while the current batch computes."""
+
+    def __init__(self, base, device, stream):
+        self.base = base
+        self.device = device
+        self.stream = stream
+        self._next = None
+        self._preload()
+
+    def _to_device(self, batch):
+        if torch.is_tensor(batch):
+            return batch.to(self.device, non_blocking=True)
+        if isinstance(batch, (list, tuple)):
+            return type(batch)(self._to_device(b) for b in batch)
+        if isinstance(batch, dict):
+            return {k: self._to_device(v) for k, v in batch.items()}
+        return batch
+
+    def _preload(self):
+        try:
+            cpu_batch = next(self.base)
+        except StopIteration:
+            self._next = None
+            return
+        with torch.cuda.stream(self.stream):
+            self._next = self._to_device(cpu_batch)
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        if self._next is None:
+            raise StopIteration
+        torch.cuda.current_stream().wait_stream(self.stream)
+        batch = self._next
+        self._preload()
+        return batch

@@ -1,0 +1,33 @@
+"""Multi-process distributed training over gloo (world_size=2, CPU)."""
+import pytest
+
+from maggy_amd import experiment
+from maggy_amd.config import TorchDistributedConfig
+from tests import _dist_fns as fns
+
+
+@pytest.mark.timeout(180)
+def test_ddp_training_gloo(exp_dir):
+    cfg = TorchDistributedConfig(
+        module=fns.TinyNet, hparams={"hidden": 16}, num_gpus=2,
+        name="ddp-gloo")
+    res = experiment.lagom(fns.dist_train_fn, cfg)
+    assert res["world_size"] == 2
+    assert res["final_metric_avg"] is not None
+    assert set(res["per_rank"].keys()) == {"0", "1"}
+
+
+@pytest.mark.timeout(180)
+def test_patched_dataloader_shards(exp_dir):
+    cfg = TorchDistributedConfig(
+        module=fns.TinyNet, num_gpus=2, name="dl-gloo")
+    res = experiment.lagom(fns.dist_dataloader_fn, cfg)
+    assert res["final_metric_avg"] == 1.0
+
+
+@pytest.mark.timeout(180)
+def test_zero_sharded_optimizer_gloo(exp_dir):
+    cfg = TorchDistributedConfig(
+        module=fns.TinyNet, num_gpus=2, zero_lvl=1, name="zero-gloo")
+    res = experiment.lagom(fns.dist_zero_fn, cfg)
+    assert res["final_metric_avg"] is not None
