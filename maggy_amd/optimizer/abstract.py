@@ -17,7 +17,7 @@ from maggy_amd.trial import Trial
 class AbstractOptimizer(ABC):
     experiment_type = "optimization"
 
-    def __init__(self):
+    def __init__(self, pruner=None, pruner_kwargs=None):
         self.searchspace = None
         self.num_trials = None
         self.trial_store = None   # dict trial_id -> running Trial
@@ -26,6 +26,25 @@ class AbstractOptimizer(ABC):
         self.pruner = None
         self.interim_results = False
         self._log_fd = None
+        if pruner:
+            self.init_pruner(pruner, pruner_kwargs or {})
+
+    def init_pruner(self, pruner, pruner_kwargs):
+        """Attach a pruner (parity abstractoptimizer.py:297-315); the metric
+        getter hands it min-convention finalized metrics."""
+        from maggy_amd.pruner import AbstractPruner, Hyperband
+
+        if isinstance(pruner, AbstractPruner):
+            self.pruner = pruner
+            if self.pruner.trial_metric_getter is None:
+                self.pruner.trial_metric_getter = self.get_metrics_dict
+        elif pruner == "hyperband":
+            self.pruner = Hyperband(
+                trial_metric_getter=self.get_metrics_dict, **pruner_kwargs)
+        else:
+            raise ValueError(
+                "Unknown pruner '{}'; expected 'hyperband' or an "
+                "AbstractPruner".format(pruner))
 
     # -- wiring (called by the driver) ----------------------------------
     def _initialize(self, exp_dir=None):

@@ -9,8 +9,8 @@ from maggy_amd.optimizer.abstract import AbstractOptimizer
 
 
 class RandomSearch(AbstractOptimizer):
-    def __init__(self):
-        super().__init__()
+    def __init__(self, pruner=None, pruner_kwargs=None):
+        super().__init__(pruner=pruner, pruner_kwargs=pruner_kwargs)
         self.config_buffer = []
 
     def initialize(self):
