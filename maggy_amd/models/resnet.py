@@ -5,12 +5,18 @@ framework).  ResNet-50/101/152 with the standard v1.5 stride placement
 (stride 2 in the 3x3 of the bottleneck).  Used by BASELINE.json configs 2/3
 (ResNet-50 synthetic 224x224 HPO/ASHA) and the samples/sec benchmark.
 
-MI355X notes: run under channels_last + bf16 autocast so MIOpen picks its
-NHWC kernels; the model is a plain nn.Module — the fused optimizer and DDP
-comm hooks live in maggy_amd.ops / maggy_amd.parallel.
+MI355X design: every ``BN -> ReLU`` and ``BN -> add -> ReLU`` pattern is a
+single MaggyBatchNorm2d module (ops/fused_bn.py) so the hand-written fused
+HIP kernels run the whole normalization/activation/residual glue in bf16
+channels_last (the eager path was 53% of the step, profiles/r01).  On CPU
+or without the extension the module falls back to torch batch_norm, so the
+architecture and parameter count are unchanged (matches torchvision's
+25,557,032 params for ResNet-50).
 """
 import torch
 import torch.nn as nn
+
+from maggy_amd.ops.fused_bn import MaggyBatchNorm2d
 
 
 class Bottleneck(nn.Module):
@@ -20,23 +26,20 @@ class Bottleneck(nn.Module):
         super().__init__()
         out_ch = width * self.expansion
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = MaggyBatchNorm2d(width, relu=True)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = MaggyBatchNorm2d(width, relu=True)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(out_ch)
-        self.relu = nn.ReLU(inplace=True)
+        # bn3 fuses the residual add + final relu
+        self.bn3 = MaggyBatchNorm2d(out_ch, relu=True)
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        return self.relu(out + identity)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=identity)
 
 
 class ResNet(nn.Module):
@@ -44,8 +47,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.in_planes = 64
         self.conv1 = nn.Conv2d(in_ch, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = MaggyBatchNorm2d(64, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0])
         self.layer2 = self._make_layer(128, layers[1], stride=2)
@@ -72,7 +74,7 @@ class ResNet(nn.Module):
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_planes, out_ch, 1, stride=stride,
                           bias=False),
-                nn.BatchNorm2d(out_ch),
+                MaggyBatchNorm2d(out_ch, relu=False),
             )
         layers = [Bottleneck(self.in_planes, width, stride, downsample)]
         self.in_planes = out_ch
@@ -81,7 +83,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x)
         return self.fc(torch.flatten(x, 1))

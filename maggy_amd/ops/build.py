@@ -18,6 +18,7 @@ SO_PATH = os.path.join(BUILD_DIR, SO_NAME + ".so")
 SOURCES = [
     os.path.join(HIP_DIR, "bindings.cpp"),
     os.path.join(HIP_DIR, "maggy_kernels.hip"),
+    os.path.join(HIP_DIR, "fused_bn.hip"),
 ]
 
 

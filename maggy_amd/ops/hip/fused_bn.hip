@@ -1,0 +1,378 @@
+// Fused BatchNorm2d (+ReLU) (+residual add) for NHWC (channels_last) bf16,
+// MI355X (gfx950, CDNA4).
+//
+// Motivation (profiles/r01): MIOpen's 4-kernel spatial BN plus the unfused
+// eager add/relu/clamp glue is 53% of a ResNet-50 bf16 step.  This file
+// collapses each BN layer-pass to:
+//   forward:  bn_sum_partial  (per-channel sum/sumsq, atomics)
+//             bn_fwd_finalize (C-thread: mean/var/running stats + scale/bias)
+//             bn_fwd_apply    (y = relu(x*scale + bias [+ residual]))
+//   backward: bn_bwd_reduce   (per-channel sum(dy_eff), sum(dy_eff*xhat),
+//                              dy_eff = relu-masked dy; atomics)
+//             bn_bwd_finalize (C-thread: dgamma/dbeta + dx coefficients)
+//             bn_bwd_apply    (dx = a*dy_eff + b + d*(x-mean) [, dres])
+//
+// Geometry: channels_last puts C innermost, so a wave reads adjacent
+// channels: 64 lanes x 2 channels (ushort2, 4 B/lane) for the reductions,
+// 8 bf16 (16 B/lane) for the elementwise passes (Guideline 13).  Per-block
+// partials reduce through LDS, one float atomicAdd per channel per block
+// (Guideline 12).  Two-pass statistics are exact (fp32 accumulation).
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define BN_THREADS 256
+#define BN_LANES 64          // lanes across channels
+#define BN_ROWS 4            // rows per block iteration (BN_THREADS/BN_LANES)
+#define BN_CHUNK (BN_LANES * 2)  // channels per block (2 per lane)
+
+__device__ __forceinline__ float bnb2f(uint16_t h) {
+  union { uint32_t u; float f; } v;
+  v.u = ((uint32_t)h) << 16;
+  return v.f;
+}
+
+__device__ __forceinline__ uint16_t bnf2b(float f) {
+  union { uint32_t u; float f; } v;
+  v.f = f;
+  uint32_t u = v.u + (0x7FFF + ((v.u >> 16) & 1));
+  return (uint16_t)(u >> 16);
+}
+
+// ---------------------------------------------------------------- forward
+
+// Per-channel sum and sum-of-squares over M = N*H*W rows.
+// grid.x = ceil(C / BN_CHUNK), grid.y = row blocks; sums/sumsqs zeroed.
+extern "C" __global__ __launch_bounds__(BN_THREADS)
+void bn_sum_partial(const uint16_t* __restrict__ x, long long M, int C,
+                    float* __restrict__ sums, float* __restrict__ sumsqs) {
+  const int lane = threadIdx.x % BN_LANES;
+  const int ry = threadIdx.x / BN_LANES;
+  const int c0 = blockIdx.x * BN_CHUNK + lane * 2;
+  float s0 = 0.f, s1 = 0.f, q0 = 0.f, q1 = 0.f;
+  if (c0 < C) {
+    const bool two = (c0 + 1) < C;
+    for (long long r = (long long)blockIdx.y * BN_ROWS + ry; r < M;
+         r += (long long)gridDim.y * BN_ROWS) {
+      const uint16_t* row = x + r * C + c0;
+      const uint32_t w = two ? *(const uint32_t*)row : (uint32_t)row[0];
+      const float a = bnb2f((uint16_t)(w & 0xFFFF));
+      const float b = bnb2f((uint16_t)(w >> 16));
+      s0 += a; q0 = fmaf(a, a, q0);
+      if (two) { s1 += b; q1 = fmaf(b, b, q1); }
+    }
+  }
+  __shared__ float lds[BN_ROWS][BN_CHUNK][2];
+  lds[ry][lane * 2][0] = s0;     lds[ry][lane * 2][1] = q0;
+  lds[ry][lane * 2 + 1][0] = s1; lds[ry][lane * 2 + 1][1] = q1;
+  __syncthreads();
+  // threads 0..BN_CHUNK-1 fold the BN_ROWS partials and push one atomic
+  if (threadIdx.x < BN_CHUNK) {
+    const int c = blockIdx.x * BN_CHUNK + threadIdx.x;
+    if (c < C) {
+      float s = 0.f, q = 0.f;
+      #pragma unroll
+      for (int i = 0; i < BN_ROWS; ++i) {
+        s += lds[i][threadIdx.x][0];
+        q += lds[i][threadIdx.x][1];
+      }
+      atomicAdd(&sums[c], s);
+      atomicAdd(&sumsqs[c], q);
+    }
+  }
+}
+
+// One block of >=C threads: statistics + running-stat update + the apply
+// coefficients scale/shift.
+extern "C" __global__
+void bn_fwd_finalize(const float* __restrict__ sums,
+                     const float* __restrict__ sumsqs, long long M, int C,
+                     const float* __restrict__ gamma,
+                     const float* __restrict__ beta,
+                     float* __restrict__ running_mean,
+                     float* __restrict__ running_var, float momentum,
+                     float eps, int training,
+                     float* __restrict__ save_mean,
+                     float* __restrict__ save_inv_std,
+                     float* __restrict__ scale,
+                     float* __restrict__ shift) {
+  for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
+       c += blockDim.x * gridDim.x) {
+    float mean, var;
+    if (training) {
+      mean = sums[c] / (float)M;
+      var = fmaxf(sumsqs[c] / (float)M - mean * mean, 0.0f);
+      // unbiased running var (torch semantics)
+      const float unbiased = M > 1 ? var * (float)M / (float)(M - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] +
+                       momentum * unbiased;
+    } else {
+      mean = running_mean[c];
+      var = running_var[c];
+    }
+    const float inv_std = rsqrtf(var + eps);
+    save_mean[c] = mean;
+    save_inv_std[c] = inv_std;
+    const float sc = gamma[c] * inv_std;
+    scale[c] = sc;
+    shift[c] = beta[c] - mean * sc;
+  }
+}
+
+// y = act(x * scale[c] + shift[c] [+ residual]); act = relu when relu != 0.
+// 8 bf16 per lane; total = M * C elements, C assumed multiple of 8 when
+// vectorizing (host falls back to the scalar tail kernel variant below via
+// the vec flag).
+extern "C" __global__ __launch_bounds__(BN_THREADS)
+void bn_fwd_apply_vec8(const uint16_t* __restrict__ x,
+                       const uint16_t* __restrict__ residual,
+                       uint16_t* __restrict__ y, long long total, int C,
+                       const float* __restrict__ scale,
+                       const float* __restrict__ shift, int relu) {
+  const long long nv = total / 8;
+  for (long long i = (long long)blockIdx.x * BN_THREADS + threadIdx.x;
+       i < nv; i += (long long)gridDim.x * BN_THREADS) {
+    const long long e = i * 8;
+    const int c = (int)(e % C);  // C % 8 == 0 -> all 8 share the base c
+    const uint4 raw = *(const uint4*)(x + e);
+    uint4 res;
+    if (residual) res = *(const uint4*)(residual + e);
+    const uint32_t w[4] = {raw.x, raw.y, raw.z, raw.w};
+    const uint32_t rw[4] = {residual ? res.x : 0, residual ? res.y : 0,
+                            residual ? res.z : 0, residual ? res.w : 0};
+    uint32_t out[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int cj = c + j * 2;
+      float a = fmaf(bnb2f((uint16_t)(w[j] & 0xFFFF)), scale[cj],
+                     shift[cj]);
+      float b = fmaf(bnb2f((uint16_t)(w[j] >> 16)), scale[cj + 1],
+                     shift[cj + 1]);
+      if (residual) {
+        a += bnb2f((uint16_t)(rw[j] & 0xFFFF));
+        b += bnb2f((uint16_t)(rw[j] >> 16));
+      }
+      if (relu) { a = fmaxf(a, 0.f); b = fmaxf(b, 0.f); }
+      out[j] = (uint32_t)bnf2b(a) | ((uint32_t)bnf2b(b) << 16);
+    }
+    *(uint4*)(y + e) = make_uint4(out[0], out[1], out[2], out[3]);
+  }
+}
+
+// ---------------------------------------------------------------- backward
+
+// Per-channel sum(dy_eff) and sum(dy_eff * xhat); dy_eff = dy masked by
+// y>0 when relu was fused.  grids like bn_sum_partial; outputs zeroed.
+extern "C" __global__ __launch_bounds__(BN_THREADS)
+void bn_bwd_reduce(const uint16_t* __restrict__ dy,
+                   const uint16_t* __restrict__ x,
+                   const uint16_t* __restrict__ y, long long M, int C,
+                   const float* __restrict__ save_mean,
+                   const float* __restrict__ save_inv_std, int relu,
+                   float* __restrict__ sum_dy,
+                   float* __restrict__ sum_dy_xhat) {
+  const int lane = threadIdx.x % BN_LANES;
+  const int ry = threadIdx.x / BN_LANES;
+  const int c0 = blockIdx.x * BN_CHUNK + lane * 2;
+  float s0 = 0.f, s1 = 0.f, t0 = 0.f, t1 = 0.f;
+  if (c0 < C) {
+    const bool two = (c0 + 1) < C;
+    const float m0 = save_mean[c0];
+    const float i0 = save_inv_std[c0];
+    const float m1 = two ? save_mean[c0 + 1] : 0.f;
+    const float i1 = two ? save_inv_std[c0 + 1] : 0.f;
+    for (long long r = (long long)blockIdx.y * BN_ROWS + ry; r < M;
+         r += (long long)gridDim.y * BN_ROWS) {
+      const long long off = r * C + c0;
+      const uint32_t wd = two ? *(const uint32_t*)(dy + off)
+                              : (uint32_t)dy[off];
+      const uint32_t wx = two ? *(const uint32_t*)(x + off)
+                              : (uint32_t)x[off];
+      float d0 = bnb2f((uint16_t)(wd & 0xFFFF));
+      float d1 = bnb2f((uint16_t)(wd >> 16));
+      if (relu) {
+        const uint32_t wy = two ? *(const uint32_t*)(y + off)
+                                : (uint32_t)y[off];
+        if (bnb2f((uint16_t)(wy & 0xFFFF)) <= 0.f) d0 = 0.f;
+        if (bnb2f((uint16_t)(wy >> 16)) <= 0.f) d1 = 0.f;
+      }
+      const float xh0 = (bnb2f((uint16_t)(wx & 0xFFFF)) - m0) * i0;
+      s0 += d0; t0 = fmaf(d0, xh0, t0);
+      if (two) {
+        const float xh1 = (bnb2f((uint16_t)(wx >> 16)) - m1) * i1;
+        s1 += d1; t1 = fmaf(d1, xh1, t1);
+      }
+    }
+  }
+  __shared__ float lds[BN_ROWS][BN_CHUNK][2];
+  lds[ry][lane * 2][0] = s0;     lds[ry][lane * 2][1] = t0;
+  lds[ry][lane * 2 + 1][0] = s1; lds[ry][lane * 2 + 1][1] = t1;
+  __syncthreads();
+  if (threadIdx.x < BN_CHUNK) {
+    const int c = blockIdx.x * BN_CHUNK + threadIdx.x;
+    if (c < C) {
+      float s = 0.f, t = 0.f;
+      #pragma unroll
+      for (int i = 0; i < BN_ROWS; ++i) {
+        s += lds[i][threadIdx.x][0];
+        t += lds[i][threadIdx.x][1];
+      }
+      atomicAdd(&sum_dy[c], s);
+      atomicAdd(&sum_dy_xhat[c], t);
+    }
+  }
+}
+
+// dgamma/dbeta + the three dx coefficients:
+//   dx = a[c]*dy_eff + b[c] + d[c]*(x - mean[c])
+//   a = gamma*inv_std, b = -a*sum_dy/M, d = -a*inv_std^2*sum_dy_xhat/M...
+// derivation: dx = a*(dy_eff - sum_dy/M - xhat*sum_dy_xhat/M),
+//   xhat = (x-mean)*inv_std  =>  d = -a*inv_std*sum_dy_xhat/M.
+extern "C" __global__
+void bn_bwd_finalize(const float* __restrict__ sum_dy,
+                     const float* __restrict__ sum_dy_xhat, long long M,
+                     int C, const float* __restrict__ gamma,
+                     const float* __restrict__ save_inv_std,
+                     float* __restrict__ dgamma, float* __restrict__ dbeta,
+                     float* __restrict__ coef_a, float* __restrict__ coef_b,
+                     float* __restrict__ coef_d) {
+  for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
+       c += blockDim.x * gridDim.x) {
+    const float inv_std = save_inv_std[c];
+    dgamma[c] = sum_dy_xhat[c];
+    dbeta[c] = sum_dy[c];
+    const float a = gamma[c] * inv_std;
+    coef_a[c] = a;
+    coef_b[c] = -a * sum_dy[c] / (float)M;
+    coef_d[c] = -a * inv_std * sum_dy_xhat[c] / (float)M;
+  }
+}
+
+// dx (+ optional residual grad = dy_eff).  8 bf16/lane like the fwd apply.
+extern "C" __global__ __launch_bounds__(BN_THREADS)
+void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
+                       const uint16_t* __restrict__ x,
+                       const uint16_t* __restrict__ y,
+                       uint16_t* __restrict__ dx,
+                       uint16_t* __restrict__ dres, long long total, int C,
+                       const float* __restrict__ save_mean,
+                       const float* __restrict__ coef_a,
+                       const float* __restrict__ coef_b,
+                       const float* __restrict__ coef_d, int relu) {
+  const long long nv = total / 8;
+  for (long long i = (long long)blockIdx.x * BN_THREADS + threadIdx.x;
+       i < nv; i += (long long)gridDim.x * BN_THREADS) {
+    const long long e = i * 8;
+    const int c = (int)(e % C);
+    const uint4 rd = *(const uint4*)(dy + e);
+    const uint4 rx = *(const uint4*)(x + e);
+    uint4 ry4;
+    if (relu) ry4 = *(const uint4*)(y + e);
+    const uint32_t wd[4] = {rd.x, rd.y, rd.z, rd.w};
+    const uint32_t wx[4] = {rx.x, rx.y, rx.z, rx.w};
+    const uint32_t wy[4] = {relu ? ry4.x : 0, relu ? ry4.y : 0,
+                            relu ? ry4.z : 0, relu ? ry4.w : 0};
+    uint32_t odx[4], odr[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int cj = c + j * 2;
+      float d0 = bnb2f((uint16_t)(wd[j] & 0xFFFF));
+      float d1 = bnb2f((uint16_t)(wd[j] >> 16));
+      if (relu) {
+        if (bnb2f((uint16_t)(wy[j] & 0xFFFF)) <= 0.f) d0 = 0.f;
+        if (bnb2f((uint16_t)(wy[j] >> 16)) <= 0.f) d1 = 0.f;
+      }
+      const float x0 = bnb2f((uint16_t)(wx[j] & 0xFFFF));
+      const float x1 = bnb2f((uint16_t)(wx[j] >> 16));
+      const float r0 = fmaf(coef_d[cj], x0 - save_mean[cj],
+                            fmaf(coef_a[cj], d0, coef_b[cj]));
+      const float r1 = fmaf(coef_d[cj + 1], x1 - save_mean[cj + 1],
+                            fmaf(coef_a[cj + 1], d1, coef_b[cj + 1]));
+      odx[j] = (uint32_t)bnf2b(r0) | ((uint32_t)bnf2b(r1) << 16);
+      if (dres)
+        odr[j] = (uint32_t)bnf2b(d0) | ((uint32_t)bnf2b(d1) << 16);
+    }
+    *(uint4*)(dx + e) = make_uint4(odx[0], odx[1], odx[2], odx[3]);
+    if (dres)
+      *(uint4*)(dres + e) = make_uint4(odr[0], odr[1], odr[2], odr[3]);
+  }
+}
+
+// ------------------------------------------------- host launch wrappers
+
+static inline dim3 bn_reduce_grid(long long M, int C) {
+  const int cx = (C + BN_CHUNK - 1) / BN_CHUNK;
+  long long rows = (M + BN_ROWS - 1) / BN_ROWS;
+  // >= 2048 blocks total to fill 256 CUs x 8 (Guideline 11)
+  long long gy = rows;
+  const long long target = (2048 + cx - 1) / cx;
+  if (gy > target) gy = target;
+  if (gy < 1) gy = 1;
+  return dim3(cx, (unsigned)gy, 1);
+}
+
+static inline int bn_elem_grid(long long total) {
+  long long g = (total / 8 + BN_THREADS - 1) / BN_THREADS;
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+extern "C" void launch_bn_fwd(
+    const void* x, const void* residual, void* y, long long M, int C,
+    const void* gamma, const void* beta, void* running_mean,
+    void* running_var, float momentum, float eps, int training, int relu,
+    void* workspace,  // 6*C floats: sums, sumsqs, save_mean, save_inv_std,
+                      // scale, shift
+    hipStream_t stream) {
+  float* ws = (float*)workspace;
+  float* sums = ws;
+  float* sumsqs = ws + C;
+  float* save_mean = ws + 2 * C;
+  float* save_inv_std = ws + 3 * C;
+  float* scale = ws + 4 * C;
+  float* shift = ws + 5 * C;
+  if (training) {
+    hipMemsetAsync(sums, 0, 2 * C * sizeof(float), stream);
+    hipLaunchKernelGGL(bn_sum_partial, bn_reduce_grid(M, C),
+                       dim3(BN_THREADS), 0, stream, (const uint16_t*)x, M,
+                       C, sums, sumsqs);
+  }
+  hipLaunchKernelGGL(bn_fwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, sums, sumsqs, M, C, (const float*)gamma,
+                     (const float*)beta, (float*)running_mean,
+                     (float*)running_var, momentum, eps, training,
+                     save_mean, save_inv_std, scale, shift);
+  hipLaunchKernelGGL(bn_fwd_apply_vec8, dim3(bn_elem_grid(M * (long long)C)),
+                     dim3(BN_THREADS), 0, stream, (const uint16_t*)x,
+                     (const uint16_t*)residual, (uint16_t*)y,
+                     M * (long long)C, C, scale, shift, relu);
+}
+
+extern "C" void launch_bn_bwd(
+    const void* dy, const void* x, const void* y, void* dx, void* dres,
+    long long M, int C, const void* gamma, const void* save_mean,
+    const void* save_inv_std, int relu, void* dgamma, void* dbeta,
+    void* workspace,  // 5*C floats: sum_dy, sum_dy_xhat, coef_a/b/d
+    hipStream_t stream) {
+  float* ws = (float*)workspace;
+  float* sum_dy = ws;
+  float* sum_dy_xhat = ws + C;
+  float* coef_a = ws + 2 * C;
+  float* coef_b = ws + 3 * C;
+  float* coef_d = ws + 4 * C;
+  hipMemsetAsync(sum_dy, 0, 2 * C * sizeof(float), stream);
+  hipLaunchKernelGGL(bn_bwd_reduce, bn_reduce_grid(M, C), dim3(BN_THREADS),
+                     0, stream, (const uint16_t*)dy, (const uint16_t*)x,
+                     (const uint16_t*)y, M, C, (const float*)save_mean,
+                     (const float*)save_inv_std, relu, sum_dy, sum_dy_xhat);
+  hipLaunchKernelGGL(bn_bwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, sum_dy, sum_dy_xhat, M, C, (const float*)gamma,
+                     (const float*)save_inv_std, (float*)dgamma,
+                     (float*)dbeta, coef_a, coef_b, coef_d);
+  hipLaunchKernelGGL(bn_bwd_apply_vec8, dim3(bn_elem_grid(M * (long long)C)),
+                     dim3(BN_THREADS), 0, stream, (const uint16_t*)dy,
+                     (const uint16_t*)x, (const uint16_t*)y, (uint16_t*)dx,
+                     (uint16_t*)dres, M * (long long)C, C,
+                     (const float*)save_mean, coef_a, coef_b, coef_d, relu);
+}
