@@ -120,20 +120,29 @@ void bn_fwd_finalize(const float* __restrict__ sums,
 }
 
 // y = act(x * scale[c] + shift[c] [+ residual]); act = relu when relu != 0.
-// 8 bf16 per lane; total = M * C elements, C assumed multiple of 8 when
-// vectorizing (host falls back to the scalar tail kernel variant below via
-// the vec flag).
+// 8 bf16 per lane.  REQUIRES the element stride (gridDim.x * BN_THREADS
+// * 8) to be a multiple of C (the host wrapper guarantees it): then every
+// thread touches the SAME 8 channels on every iteration, so the per-channel
+// coefficients load into registers once — the previous per-iteration gather
+// of 16 scalar loads made this kernel issue-bound (profiles/r02).
 extern "C" __global__ __launch_bounds__(BN_THREADS)
 void bn_fwd_apply_vec8(const uint16_t* __restrict__ x,
                        const uint16_t* __restrict__ residual,
                        uint16_t* __restrict__ y, long long total, int C,
                        const float* __restrict__ scale,
                        const float* __restrict__ shift, int relu) {
-  const long long nv = total / 8;
-  for (long long i = (long long)blockIdx.x * BN_THREADS + threadIdx.x;
-       i < nv; i += (long long)gridDim.x * BN_THREADS) {
-    const long long e = i * 8;
-    const int c = (int)(e % C);  // C % 8 == 0 -> all 8 share the base c
+  const long long e0 =
+      ((long long)blockIdx.x * BN_THREADS + threadIdx.x) * 8;
+  if (e0 >= total) return;
+  const long long stride = (long long)gridDim.x * BN_THREADS * 8;
+  const int c = (int)(e0 % C);
+  float sc[8], sh[8];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = scale[c + j];
+    sh[j] = shift[c + j];
+  }
+  for (long long e = e0; e < total; e += stride) {
     const uint4 raw = *(const uint4*)(x + e);
     uint4 res;
     if (residual) res = *(const uint4*)(residual + e);
@@ -143,11 +152,10 @@ void bn_fwd_apply_vec8(const uint16_t* __restrict__ x,
     uint32_t out[4];
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const int cj = c + j * 2;
-      float a = fmaf(bnb2f((uint16_t)(w[j] & 0xFFFF)), scale[cj],
-                     shift[cj]);
-      float b = fmaf(bnb2f((uint16_t)(w[j] >> 16)), scale[cj + 1],
-                     shift[cj + 1]);
+      float a = fmaf(bnb2f((uint16_t)(w[j] & 0xFFFF)), sc[j * 2],
+                     sh[j * 2]);
+      float b = fmaf(bnb2f((uint16_t)(w[j] >> 16)), sc[j * 2 + 1],
+                     sh[j * 2 + 1]);
       if (residual) {
         a += bnb2f((uint16_t)(rw[j] & 0xFFFF));
         b += bnb2f((uint16_t)(rw[j] >> 16));
@@ -248,7 +256,9 @@ void bn_bwd_finalize(const float* __restrict__ sum_dy,
   }
 }
 
-// dx (+ optional residual grad = dy_eff).  8 bf16/lane like the fwd apply.
+// dx (+ optional residual grad = dy_eff).  8 bf16/lane; same fixed-channel
+// stride contract as bn_fwd_apply_vec8 — all per-channel coefficients live
+// in registers across the grid-stride loop.
 extern "C" __global__ __launch_bounds__(BN_THREADS)
 void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
                        const uint16_t* __restrict__ x,
@@ -259,11 +269,20 @@ void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
                        const float* __restrict__ coef_a,
                        const float* __restrict__ coef_b,
                        const float* __restrict__ coef_d, int relu) {
-  const long long nv = total / 8;
-  for (long long i = (long long)blockIdx.x * BN_THREADS + threadIdx.x;
-       i < nv; i += (long long)gridDim.x * BN_THREADS) {
-    const long long e = i * 8;
-    const int c = (int)(e % C);
+  const long long e0 =
+      ((long long)blockIdx.x * BN_THREADS + threadIdx.x) * 8;
+  if (e0 >= total) return;
+  const long long stride = (long long)gridDim.x * BN_THREADS * 8;
+  const int c = (int)(e0 % C);
+  float mu[8], ca[8], cb[8], cd[8];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mu[j] = save_mean[c + j];
+    ca[j] = coef_a[c + j];
+    cb[j] = coef_b[c + j];
+    cd[j] = coef_d[c + j];
+  }
+  for (long long e = e0; e < total; e += stride) {
     const uint4 rd = *(const uint4*)(dy + e);
     const uint4 rx = *(const uint4*)(x + e);
     uint4 ry4;
@@ -275,7 +294,6 @@ void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
     uint32_t odx[4], odr[4];
     #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const int cj = c + j * 2;
       float d0 = bnb2f((uint16_t)(wd[j] & 0xFFFF));
       float d1 = bnb2f((uint16_t)(wd[j] >> 16));
       if (relu) {
@@ -284,10 +302,10 @@ void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
       }
       const float x0 = bnb2f((uint16_t)(wx[j] & 0xFFFF));
       const float x1 = bnb2f((uint16_t)(wx[j] >> 16));
-      const float r0 = fmaf(coef_d[cj], x0 - save_mean[cj],
-                            fmaf(coef_a[cj], d0, coef_b[cj]));
-      const float r1 = fmaf(coef_d[cj + 1], x1 - save_mean[cj + 1],
-                            fmaf(coef_a[cj + 1], d1, coef_b[cj + 1]));
+      const float r0 = fmaf(cd[j * 2], x0 - mu[j * 2],
+                            fmaf(ca[j * 2], d0, cb[j * 2]));
+      const float r1 = fmaf(cd[j * 2 + 1], x1 - mu[j * 2 + 1],
+                            fmaf(ca[j * 2 + 1], d1, cb[j * 2 + 1]));
       odx[j] = (uint32_t)bnf2b(r0) | ((uint32_t)bnf2b(r1) << 16);
       if (dres)
         odr[j] = (uint32_t)bnf2b(d0) | ((uint32_t)bnf2b(d1) << 16);
@@ -311,10 +329,21 @@ static inline dim3 bn_reduce_grid(long long M, int C) {
   return dim3(cx, (unsigned)gy, 1);
 }
 
-static inline int bn_elem_grid(long long total) {
+static inline long long bn_gcd(long long a, long long b) {
+  while (b) { long long t = a % b; a = b; b = t; }
+  return a;
+}
+
+// grid for the vec8 apply kernels; the element stride grid*BN_THREADS*8
+// must be a multiple of C (fixed-channel contract).  For ResNet's
+// power-of-2 channel counts any grid works; general C rounds up to a
+// multiple of C/gcd(C, 2048).
+static inline int bn_elem_grid(long long total, int C) {
   long long g = (total / 8 + BN_THREADS - 1) / BN_THREADS;
   if (g > 2048) g = 2048;
   if (g < 1) g = 1;
+  const long long m = C / bn_gcd((long long)C, (long long)BN_THREADS * 8);
+  g = ((g + m - 1) / m) * m;
   return (int)g;
 }
 
@@ -343,7 +372,7 @@ extern "C" void launch_bn_fwd(
                      (const float*)beta, (float*)running_mean,
                      (float*)running_var, momentum, eps, training,
                      save_mean, save_inv_std, scale, shift);
-  hipLaunchKernelGGL(bn_fwd_apply_vec8, dim3(bn_elem_grid(M * (long long)C)),
+  hipLaunchKernelGGL(bn_fwd_apply_vec8, dim3(bn_elem_grid(M * (long long)C, C)),
                      dim3(BN_THREADS), 0, stream, (const uint16_t*)x,
                      (const uint16_t*)residual, (uint16_t*)y,
                      M * (long long)C, C, scale, shift, relu);
@@ -370,7 +399,7 @@ extern "C" void launch_bn_bwd(
                      stream, sum_dy, sum_dy_xhat, M, C, (const float*)gamma,
                      (const float*)save_inv_std, (float*)dgamma,
                      (float*)dbeta, coef_a, coef_b, coef_d);
-  hipLaunchKernelGGL(bn_bwd_apply_vec8, dim3(bn_elem_grid(M * (long long)C)),
+  hipLaunchKernelGGL(bn_bwd_apply_vec8, dim3(bn_elem_grid(M * (long long)C, C)),
                      dim3(BN_THREADS), 0, stream, (const uint16_t*)dy,
                      (const uint16_t*)x, (const uint16_t*)y, (uint16_t*)dx,
                      (uint16_t*)dres, M * (long long)C, C,
