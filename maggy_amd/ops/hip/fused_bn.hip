@@ -40,45 +40,74 @@ __device__ __forceinline__ uint16_t bnf2b(float f) {
 
 // ---------------------------------------------------------------- forward
 
-// Per-channel sum and sum-of-squares over M = N*H*W rows.
-// grid.x = ceil(C / BN_CHUNK), grid.y = row blocks; sums/sumsqs zeroed.
-extern "C" __global__ __launch_bounds__(BN_THREADS)
-void bn_sum_partial(const uint16_t* __restrict__ x, long long M, int C,
-                    float* __restrict__ sums, float* __restrict__ sumsqs) {
-  const int lane = threadIdx.x % BN_LANES;
-  const int ry = threadIdx.x / BN_LANES;
-  const int c0 = blockIdx.x * BN_CHUNK + lane * 2;
-  float s0 = 0.f, s1 = 0.f, q0 = 0.f, q1 = 0.f;
-  if (c0 < C) {
-    const bool two = (c0 + 1) < C;
-    for (long long r = (long long)blockIdx.y * BN_ROWS + ry; r < M;
-         r += (long long)gridDim.y * BN_ROWS) {
-      const uint16_t* row = x + r * C + c0;
-      const uint32_t w = two ? *(const uint32_t*)row : (uint32_t)row[0];
-      const float a = bnb2f((uint16_t)(w & 0xFFFF));
-      const float b = bnb2f((uint16_t)(w >> 16));
-      s0 += a; q0 = fmaf(a, a, q0);
-      if (two) { s1 += b; q1 = fmaf(b, b, q1); }
+// LDS tree-fold of two 8-wide per-thread accumulator vectors across the
+// thread groups that share a channel set (group stride G = C/8, power of
+// 2), then C atomics per block.  acc0/acc1 are the calling thread's 8-wide
+// partials; out0/out1 the global per-channel arrays.
+__device__ __forceinline__ void bn_block_fold8(
+    float acc0[8], float acc1[8], int C, int c,
+    float* __restrict__ out0, float* __restrict__ out1) {
+  __shared__ float lds0[BN_THREADS][8];
+  __shared__ float lds1[BN_THREADS][8];
+  const int t = threadIdx.x;
+  const int G = C / 8;  // distinct channel sets per block
+  if ((G & (G - 1)) == 0 && G <= BN_THREADS) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      lds0[t][j] = acc0[j];
+      lds1[t][j] = acc1[j];
     }
-  }
-  __shared__ float lds[BN_ROWS][BN_CHUNK][2];
-  lds[ry][lane * 2][0] = s0;     lds[ry][lane * 2][1] = q0;
-  lds[ry][lane * 2 + 1][0] = s1; lds[ry][lane * 2 + 1][1] = q1;
-  __syncthreads();
-  // threads 0..BN_CHUNK-1 fold the BN_ROWS partials and push one atomic
-  if (threadIdx.x < BN_CHUNK) {
-    const int c = blockIdx.x * BN_CHUNK + threadIdx.x;
-    if (c < C) {
-      float s = 0.f, q = 0.f;
-      #pragma unroll
-      for (int i = 0; i < BN_ROWS; ++i) {
-        s += lds[i][threadIdx.x][0];
-        q += lds[i][threadIdx.x][1];
+    __syncthreads();
+    for (int s = BN_THREADS / 2; s >= G; s >>= 1) {
+      if (t < s) {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          lds0[t][j] += lds0[t + s][j];
+          lds1[t][j] += lds1[t + s][j];
+        }
       }
-      atomicAdd(&sums[c], s);
-      atomicAdd(&sumsqs[c], q);
+      __syncthreads();
+    }
+    if (t < G) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        atomicAdd(&out0[t * 8 + j], lds0[t][j]);
+        atomicAdd(&out1[t * 8 + j], lds1[t][j]);
+      }
+    }
+  } else {
+    // general C: one atomic per accumulator
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&out0[c + j], acc0[j]);
+      atomicAdd(&out1[c + j], acc1[j]);
     }
   }
+}
+
+// Per-channel sum and sum-of-squares over the flat tensor (16 B/lane,
+// fixed-channel stride contract like the apply kernels); outputs zeroed.
+extern "C" __global__ __launch_bounds__(BN_THREADS)
+void bn_sum_partial(const uint16_t* __restrict__ x, long long total, int C,
+                    float* __restrict__ sums, float* __restrict__ sumsqs) {
+  const long long e0 =
+      ((long long)blockIdx.x * BN_THREADS + threadIdx.x) * 8;
+  const long long stride = (long long)gridDim.x * BN_THREADS * 8;
+  const int c = (int)(e0 % C);
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (long long e = e0; e < total; e += stride) {
+    const uint4 raw = *(const uint4*)(x + e);
+    const uint32_t w[4] = {raw.x, raw.y, raw.z, raw.w};
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float a = bnb2f((uint16_t)(w[j] & 0xFFFF));
+      const float b = bnb2f((uint16_t)(w[j] >> 16));
+      s[j * 2] += a;     q[j * 2] = fmaf(a, a, q[j * 2]);
+      s[j * 2 + 1] += b; q[j * 2 + 1] = fmaf(b, b, q[j * 2 + 1]);
+    }
+  }
+  bn_block_fold8(s, q, C, c, sums, sumsqs);
 }
 
 // One block of >=C threads: statistics + running-stat update + the apply
@@ -170,65 +199,54 @@ void bn_fwd_apply_vec8(const uint16_t* __restrict__ x,
 // ---------------------------------------------------------------- backward
 
 // Per-channel sum(dy_eff) and sum(dy_eff * xhat); dy_eff = dy masked by
-// y>0 when relu was fused.  grids like bn_sum_partial; outputs zeroed.
+// y>0 when relu was fused.  Flat 16 B/lane layout with fixed channels per
+// thread (same stride contract as the apply kernels); outputs zeroed.
 extern "C" __global__ __launch_bounds__(BN_THREADS)
 void bn_bwd_reduce(const uint16_t* __restrict__ dy,
                    const uint16_t* __restrict__ x,
-                   const uint16_t* __restrict__ y, long long M, int C,
+                   const uint16_t* __restrict__ y, long long total, int C,
                    const float* __restrict__ save_mean,
                    const float* __restrict__ save_inv_std, int relu,
                    float* __restrict__ sum_dy,
                    float* __restrict__ sum_dy_xhat) {
-  const int lane = threadIdx.x % BN_LANES;
-  const int ry = threadIdx.x / BN_LANES;
-  const int c0 = blockIdx.x * BN_CHUNK + lane * 2;
-  float s0 = 0.f, s1 = 0.f, t0 = 0.f, t1 = 0.f;
-  if (c0 < C) {
-    const bool two = (c0 + 1) < C;
-    const float m0 = save_mean[c0];
-    const float i0 = save_inv_std[c0];
-    const float m1 = two ? save_mean[c0 + 1] : 0.f;
-    const float i1 = two ? save_inv_std[c0 + 1] : 0.f;
-    for (long long r = (long long)blockIdx.y * BN_ROWS + ry; r < M;
-         r += (long long)gridDim.y * BN_ROWS) {
-      const long long off = r * C + c0;
-      const uint32_t wd = two ? *(const uint32_t*)(dy + off)
-                              : (uint32_t)dy[off];
-      const uint32_t wx = two ? *(const uint32_t*)(x + off)
-                              : (uint32_t)x[off];
-      float d0 = bnb2f((uint16_t)(wd & 0xFFFF));
-      float d1 = bnb2f((uint16_t)(wd >> 16));
+  const long long e0 =
+      ((long long)blockIdx.x * BN_THREADS + threadIdx.x) * 8;
+  const long long stride = (long long)gridDim.x * BN_THREADS * 8;
+  const int c = (int)(e0 % C);
+  float mu[8], is[8];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mu[j] = save_mean[c + j];
+    is[j] = save_inv_std[c + j];
+  }
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float t[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (long long e = e0; e < total; e += stride) {
+    const uint4 rd = *(const uint4*)(dy + e);
+    const uint4 rx = *(const uint4*)(x + e);
+    uint4 ry4;
+    if (relu) ry4 = *(const uint4*)(y + e);
+    const uint32_t wd[4] = {rd.x, rd.y, rd.z, rd.w};
+    const uint32_t wx[4] = {rx.x, rx.y, rx.z, rx.w};
+    const uint32_t wy[4] = {relu ? ry4.x : 0, relu ? ry4.y : 0,
+                            relu ? ry4.z : 0, relu ? ry4.w : 0};
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float d0 = bnb2f((uint16_t)(wd[j] & 0xFFFF));
+      float d1 = bnb2f((uint16_t)(wd[j] >> 16));
       if (relu) {
-        const uint32_t wy = two ? *(const uint32_t*)(y + off)
-                                : (uint32_t)y[off];
-        if (bnb2f((uint16_t)(wy & 0xFFFF)) <= 0.f) d0 = 0.f;
-        if (bnb2f((uint16_t)(wy >> 16)) <= 0.f) d1 = 0.f;
+        if (bnb2f((uint16_t)(wy[j] & 0xFFFF)) <= 0.f) d0 = 0.f;
+        if (bnb2f((uint16_t)(wy[j] >> 16)) <= 0.f) d1 = 0.f;
       }
-      const float xh0 = (bnb2f((uint16_t)(wx & 0xFFFF)) - m0) * i0;
-      s0 += d0; t0 = fmaf(d0, xh0, t0);
-      if (two) {
-        const float xh1 = (bnb2f((uint16_t)(wx >> 16)) - m1) * i1;
-        s1 += d1; t1 = fmaf(d1, xh1, t1);
-      }
+      const float xh0 =
+          (bnb2f((uint16_t)(wx[j] & 0xFFFF)) - mu[j * 2]) * is[j * 2];
+      const float xh1 =
+          (bnb2f((uint16_t)(wx[j] >> 16)) - mu[j * 2 + 1]) * is[j * 2 + 1];
+      s[j * 2] += d0;     t[j * 2] = fmaf(d0, xh0, t[j * 2]);
+      s[j * 2 + 1] += d1; t[j * 2 + 1] = fmaf(d1, xh1, t[j * 2 + 1]);
     }
   }
-  __shared__ float lds[BN_ROWS][BN_CHUNK][2];
-  lds[ry][lane * 2][0] = s0;     lds[ry][lane * 2][1] = t0;
-  lds[ry][lane * 2 + 1][0] = s1; lds[ry][lane * 2 + 1][1] = t1;
-  __syncthreads();
-  if (threadIdx.x < BN_CHUNK) {
-    const int c = blockIdx.x * BN_CHUNK + threadIdx.x;
-    if (c < C) {
-      float s = 0.f, t = 0.f;
-      #pragma unroll
-      for (int i = 0; i < BN_ROWS; ++i) {
-        s += lds[i][threadIdx.x][0];
-        t += lds[i][threadIdx.x][1];
-      }
-      atomicAdd(&sum_dy[c], s);
-      atomicAdd(&sum_dy_xhat[c], t);
-    }
-  }
+  bn_block_fold8(s, t, C, c, sum_dy, sum_dy_xhat);
 }
 
 // dgamma/dbeta + the three dx coefficients:
@@ -362,10 +380,11 @@ extern "C" void launch_bn_fwd(
   float* scale = ws + 4 * C;
   float* shift = ws + 5 * C;
   if (training) {
-    hipMemsetAsync(sums, 0, 2 * C * sizeof(float), stream);
-    hipLaunchKernelGGL(bn_sum_partial, bn_reduce_grid(M, C),
-                       dim3(BN_THREADS), 0, stream, (const uint16_t*)x, M,
-                       C, sums, sumsqs);
+    (void)hipMemsetAsync(sums, 0, 2 * C * sizeof(float), stream);
+    hipLaunchKernelGGL(bn_sum_partial,
+                       dim3(bn_elem_grid(M * (long long)C, C)),
+                       dim3(BN_THREADS), 0, stream, (const uint16_t*)x,
+                       M * (long long)C, C, sums, sumsqs);
   }
   hipLaunchKernelGGL(bn_fwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
                      stream, sums, sumsqs, M, C, (const float*)gamma,
@@ -390,10 +409,12 @@ extern "C" void launch_bn_bwd(
   float* coef_a = ws + 2 * C;
   float* coef_b = ws + 3 * C;
   float* coef_d = ws + 4 * C;
-  hipMemsetAsync(sum_dy, 0, 2 * C * sizeof(float), stream);
-  hipLaunchKernelGGL(bn_bwd_reduce, bn_reduce_grid(M, C), dim3(BN_THREADS),
-                     0, stream, (const uint16_t*)dy, (const uint16_t*)x,
-                     (const uint16_t*)y, M, C, (const float*)save_mean,
+  (void)hipMemsetAsync(sum_dy, 0, 2 * C * sizeof(float), stream);
+  hipLaunchKernelGGL(bn_bwd_reduce,
+                     dim3(bn_elem_grid(M * (long long)C, C)),
+                     dim3(BN_THREADS), 0, stream, (const uint16_t*)dy,
+                     (const uint16_t*)x, (const uint16_t*)y,
+                     M * (long long)C, C, (const float*)save_mean,
                      (const float*)save_inv_std, relu, sum_dy, sum_dy_xhat);
   hipLaunchKernelGGL(bn_bwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
                      stream, sum_dy, sum_dy_xhat, M, C, (const float*)gamma,
