@@ -17,6 +17,8 @@ from maggy_amd import ops
 
 
 class _FusedBNFunction(torch.autograd.Function):
+    BN_NB = 1024  # partial rows, must match BN_NB in fused_bn.hip
+
     @staticmethod
     def forward(ctx, x, residual, weight, bias, running_mean, running_var,
                 training, momentum, eps, relu):
@@ -25,8 +27,12 @@ class _FusedBNFunction(torch.autograd.Function):
         M = N * H * W
         y = torch.empty_like(x)
         ws = torch.empty(6 * C, dtype=torch.float32, device=x.device)
+        partials = None
+        if training:
+            partials = torch.empty(2 * C * _FusedBNFunction.BN_NB,
+                                   dtype=torch.float32, device=x.device)
         ext.bn_fwd(x, residual, y, M, C, weight, bias, running_mean,
-                   running_var, momentum, eps, training, relu, ws)
+                   running_var, momentum, eps, training, relu, ws, partials)
         ctx.save_for_backward(x, y, weight, ws)
         ctx.bn_shape = (M, C)
         ctx.relu = relu
@@ -44,8 +50,10 @@ class _FusedBNFunction(torch.autograd.Function):
         dgamma = torch.empty(C, dtype=torch.float32, device=dy.device)
         dbeta = torch.empty(C, dtype=torch.float32, device=dy.device)
         bwd_ws = torch.empty(5 * C, dtype=torch.float32, device=dy.device)
+        partials = torch.empty(2 * C * _FusedBNFunction.BN_NB,
+                               dtype=torch.float32, device=dy.device)
         ext.bn_bwd(dy, x, y, dx, dres, M, C, weight, ws, ctx.relu, dgamma,
-                   dbeta, bwd_ws)
+                   dbeta, bwd_ws, partials)
         return (dx, dres, dgamma, dbeta, None, None, None, None, None, None)
 
 
@@ -61,12 +69,18 @@ class MaggyBatchNorm2d(nn.BatchNorm2d):
                          affine=True, track_running_stats=True)
         self.relu = relu
 
+    @staticmethod
+    def _eligible_channels(C):
+        # kernels require power-of-2 C in [8, 2048] (fixed-channel stride
+        # and LDS fold contracts)
+        return 8 <= C <= 2048 and (C & (C - 1)) == 0
+
     def _use_fused(self, x):
         return (
             x.is_cuda
             and x.dtype == torch.bfloat16
             and x.dim() == 4
-            and x.shape[1] % 8 == 0
+            and self._eligible_channels(x.shape[1])
             and x.is_contiguous(memory_format=torch.channels_last)
             and ops.has_ext()
         )
@@ -81,7 +95,8 @@ class MaggyBatchNorm2d(nn.BatchNorm2d):
                 self.running_var, self.training, self.momentum, self.eps,
                 self.relu)
         if (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4
-                and x.shape[1] % 8 == 0 and not ops.has_ext()):
+                and self._eligible_channels(x.shape[1])
+                and not ops.has_ext()):
             # eligible input but extension missing: fail loudly, no silent
             # eager fallback on a GPU box (framework policy)
             ops.require_ext()

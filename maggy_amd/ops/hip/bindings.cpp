@@ -19,11 +19,12 @@ extern "C" void launch_multi_fused_sgd(const void*, int, const void*, int,
                                        hipStream_t);
 extern "C" void launch_bn_fwd(const void*, const void*, void*, long long,
                               int, const void*, const void*, void*, void*,
-                              float, float, int, int, void*, hipStream_t);
+                              float, float, int, int, void*, void*,
+                              hipStream_t);
 extern "C" void launch_bn_bwd(const void*, const void*, const void*, void*,
                               void*, long long, int, const void*,
                               const void*, const void*, int, void*, void*,
-                              void*, hipStream_t);
+                              void*, void*, hipStream_t);
 extern "C" void launch_reduce_sum_f32(const float*, long long, float*,
                                       hipStream_t);
 extern "C" void launch_reduce_sum_bf16(const void*, long long, float*,
@@ -90,35 +91,47 @@ void multi_fused_sgd(torch::Tensor chunks, int64_t n_chunks,
                          (float)grad_scale_inv, current_stream());
 }
 
+constexpr int64_t kBnNB = 1024;  // must match BN_NB in fused_bn.hip
+
 void bn_fwd(torch::Tensor x, c10::optional<torch::Tensor> residual,
             torch::Tensor y, int64_t M, int64_t C, torch::Tensor gamma,
             torch::Tensor beta, torch::Tensor running_mean,
             torch::Tensor running_var, double momentum, double eps,
-            bool training, bool relu, torch::Tensor workspace) {
+            bool training, bool relu, torch::Tensor workspace,
+            c10::optional<torch::Tensor> partials) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
-  TORCH_CHECK(C % 8 == 0, "fused BN requires C % 8 == 0");
+  TORCH_CHECK(C >= 8 && C % 8 == 0 && (C & (C - 1)) == 0 && C <= 2048,
+              "fused BN requires power-of-2 C in [8, 2048]");
   TORCH_CHECK(workspace.numel() >= 6 * C);
+  void* part = nullptr;
+  if (training) {
+    TORCH_CHECK(partials.has_value() &&
+                partials->numel() >= 2 * C * kBnNB);
+    part = partials->data_ptr();
+  }
   const void* res = residual.has_value() ? residual->data_ptr() : nullptr;
   launch_bn_fwd(x.data_ptr(), res, y.data_ptr(), M, (int)C,
                 gamma.data_ptr(), beta.data_ptr(), running_mean.data_ptr(),
                 running_var.data_ptr(), (float)momentum, (float)eps,
                 training ? 1 : 0, relu ? 1 : 0, workspace.data_ptr(),
-                current_stream());
+                part, current_stream());
 }
 
 void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor y,
             torch::Tensor dx, c10::optional<torch::Tensor> dres, int64_t M,
             int64_t C, torch::Tensor gamma, torch::Tensor fwd_ws,
             bool relu, torch::Tensor dgamma, torch::Tensor dbeta,
-            torch::Tensor bwd_ws) {
+            torch::Tensor bwd_ws, torch::Tensor partials) {
   TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16);
   TORCH_CHECK(bwd_ws.numel() >= 5 * C);
+  TORCH_CHECK(partials.numel() >= 2 * C * kBnNB);
   const float* ws = fwd_ws.data_ptr<float>();
   void* dres_p = dres.has_value() ? dres->data_ptr() : nullptr;
   launch_bn_bwd(dy.data_ptr(), x.data_ptr(), y.data_ptr(), dx.data_ptr(),
                 dres_p, M, (int)C, gamma.data_ptr(), (const void*)(ws + 2 * C),
                 (const void*)(ws + 3 * C), relu ? 1 : 0, dgamma.data_ptr(),
-                dbeta.data_ptr(), bwd_ws.data_ptr(), current_stream());
+                dbeta.data_ptr(), bwd_ws.data_ptr(), partials.data_ptr(),
+                current_stream());
 }
 
 void reduce_sum(torch::Tensor in, torch::Tensor out) {

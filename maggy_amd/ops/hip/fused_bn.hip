@@ -42,15 +42,18 @@ __device__ __forceinline__ uint16_t bnf2b(float f) {
 
 // LDS tree-fold of two 8-wide per-thread accumulator vectors across the
 // thread groups that share a channel set (group stride G = C/8, power of
-// 2), then C atomics per block.  acc0/acc1 are the calling thread's 8-wide
-// partials; out0/out1 the global per-channel arrays.
+// 2), then ONE partial row per block: p[c * NB + blockIdx.x].  Global
+// atomics are NOT used — per-address atomic serialization ran at
+// 32-930 GB/s vs 6.2 TB/s for this scheme (scripts/bnrepro.hip).  A second
+// kernel (bn_fold_partials) folds the NB partials per channel.
 __device__ __forceinline__ void bn_block_fold8(
-    float acc0[8], float acc1[8], int C, int c,
-    float* __restrict__ out0, float* __restrict__ out1) {
+    float acc0[8], float acc1[8], int C, int c, int NB,
+    float* __restrict__ p0, float* __restrict__ p1) {
   __shared__ float lds0[BN_THREADS][8];
   __shared__ float lds1[BN_THREADS][8];
   const int t = threadIdx.x;
   const int G = C / 8;  // distinct channel sets per block
+  const long long b = blockIdx.x;
   if ((G & (G - 1)) == 0 && G <= BN_THREADS) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -71,25 +74,65 @@ __device__ __forceinline__ void bn_block_fold8(
     if (t < G) {
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        atomicAdd(&out0[t * 8 + j], lds0[t][j]);
-        atomicAdd(&out1[t * 8 + j], lds1[t][j]);
+        p0[(long long)(t * 8 + j) * NB + b] = lds0[t][j];
+        p1[(long long)(t * 8 + j) * NB + b] = lds1[t][j];
       }
     }
-  } else {
-    // general C: one atomic per accumulator
+  }
+  // non-power-of-2 C never reaches the fused path (gated in
+  // ops/fused_bn.py::_use_fused); (void)c silences unused in that branch
+  (void)c;
+}
+
+// wave64 + LDS block sum used by bn_fold_partials
+__device__ __forceinline__ float block_reduce_sum2(float val) {
+  for (int off = 32; off > 0; off >>= 1)
+    val += __shfl_down(val, off, 64);
+  __shared__ float lds[BN_THREADS / 64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) lds[wave] = val;
+  __syncthreads();
+  float out = 0.f;
+  if (threadIdx.x == 0) {
     #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      atomicAdd(&out0[c + j], acc0[j]);
-      atomicAdd(&out1[c + j], acc1[j]);
-    }
+    for (int i = 0; i < BN_THREADS / 64; ++i) out += lds[i];
+  }
+  __syncthreads();
+  return out;
+}
+
+// Fold NB partials per channel: block c sums p[c*NB .. c*NB+NB) (float4
+// coalesced: 256 threads x float4 covers NB=1024 in one load).
+extern "C" __global__ __launch_bounds__(BN_THREADS)
+void bn_fold_partials(const float* __restrict__ p0,
+                      const float* __restrict__ p1, int NB, int C,
+                      float* __restrict__ out0, float* __restrict__ out1) {
+  const int c = blockIdx.x;
+  float s0 = 0.f, s1 = 0.f;
+  const long long base = (long long)c * NB;
+  for (int i = threadIdx.x * 4; i < NB; i += BN_THREADS * 4) {
+    const float4 a = *(const float4*)(p0 + base + i);
+    const float4 b = *(const float4*)(p1 + base + i);
+    s0 += a.x + a.y + a.z + a.w;
+    s1 += b.x + b.y + b.z + b.w;
+  }
+  s0 = block_reduce_sum2(s0);
+  s1 = block_reduce_sum2(s1);
+  if (threadIdx.x == 0) {
+    out0[c] = s0;
+    out1[c] = s1;
   }
 }
 
 // Per-channel sum and sum-of-squares over the flat tensor (16 B/lane,
-// fixed-channel stride contract like the apply kernels); outputs zeroed.
+// fixed-channel stride contract like the apply kernels).  Writes one
+// partial row per block into p0/p1 ([C][NB] layout); MUST be launched
+// with gridDim.x == NB (idle blocks still write their zero partials).
 extern "C" __global__ __launch_bounds__(BN_THREADS)
 void bn_sum_partial(const uint16_t* __restrict__ x, long long total, int C,
-                    float* __restrict__ sums, float* __restrict__ sumsqs) {
+                    int NB, float* __restrict__ p0,
+                    float* __restrict__ p1) {
   const long long e0 =
       ((long long)blockIdx.x * BN_THREADS + threadIdx.x) * 8;
   const long long stride = (long long)gridDim.x * BN_THREADS * 8;
@@ -107,7 +150,7 @@ void bn_sum_partial(const uint16_t* __restrict__ x, long long total, int C,
       s[j * 2 + 1] += b; q[j * 2 + 1] = fmaf(b, b, q[j * 2 + 1]);
     }
   }
-  bn_block_fold8(s, q, C, c, sums, sumsqs);
+  bn_block_fold8(s, q, C, c, NB, p0, p1);
 }
 
 // One block of >=C threads: statistics + running-stat update + the apply
@@ -207,8 +250,8 @@ void bn_bwd_reduce(const uint16_t* __restrict__ dy,
                    const uint16_t* __restrict__ y, long long total, int C,
                    const float* __restrict__ save_mean,
                    const float* __restrict__ save_inv_std, int relu,
-                   float* __restrict__ sum_dy,
-                   float* __restrict__ sum_dy_xhat) {
+                   int NB, float* __restrict__ p0,
+                   float* __restrict__ p1) {
   const long long e0 =
       ((long long)blockIdx.x * BN_THREADS + threadIdx.x) * 8;
   const long long stride = (long long)gridDim.x * BN_THREADS * 8;
@@ -246,7 +289,7 @@ void bn_bwd_reduce(const uint16_t* __restrict__ dy,
       s[j * 2 + 1] += d1; t[j * 2 + 1] = fmaf(d1, xh1, t[j * 2 + 1]);
     }
   }
-  bn_block_fold8(s, t, C, c, sum_dy, sum_dy_xhat);
+  bn_block_fold8(s, t, C, c, NB, p0, p1);
 }
 
 // dgamma/dbeta + the three dx coefficients:
@@ -336,16 +379,9 @@ void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
 
 // ------------------------------------------------- host launch wrappers
 
-static inline dim3 bn_reduce_grid(long long M, int C) {
-  const int cx = (C + BN_CHUNK - 1) / BN_CHUNK;
-  long long rows = (M + BN_ROWS - 1) / BN_ROWS;
-  // >= 2048 blocks total to fill 256 CUs x 8 (Guideline 11)
-  long long gy = rows;
-  const long long target = (2048 + cx - 1) / cx;
-  if (gy > target) gy = target;
-  if (gy < 1) gy = 1;
-  return dim3(cx, (unsigned)gy, 1);
-}
+// partial-row count for the two-stage reduction: fixed power of 2 so the
+// element stride NB*BN_THREADS*8 divides by every power-of-2 C <= 2048
+#define BN_NB 1024
 
 static inline long long bn_gcd(long long a, long long b) {
   while (b) { long long t = a % b; a = b; b = t; }
@@ -371,6 +407,7 @@ extern "C" void launch_bn_fwd(
     void* running_var, float momentum, float eps, int training, int relu,
     void* workspace,  // 6*C floats: sums, sumsqs, save_mean, save_inv_std,
                       // scale, shift
+    void* partials,   // 2 * C * BN_NB floats (training only)
     hipStream_t stream) {
   float* ws = (float*)workspace;
   float* sums = ws;
@@ -380,11 +417,13 @@ extern "C" void launch_bn_fwd(
   float* scale = ws + 4 * C;
   float* shift = ws + 5 * C;
   if (training) {
-    (void)hipMemsetAsync(sums, 0, 2 * C * sizeof(float), stream);
-    hipLaunchKernelGGL(bn_sum_partial,
-                       dim3(bn_elem_grid(M * (long long)C, C)),
-                       dim3(BN_THREADS), 0, stream, (const uint16_t*)x,
-                       M * (long long)C, C, sums, sumsqs);
+    float* p0 = (float*)partials;
+    float* p1 = p0 + (long long)C * BN_NB;
+    hipLaunchKernelGGL(bn_sum_partial, dim3(BN_NB), dim3(BN_THREADS), 0,
+                       stream, (const uint16_t*)x, M * (long long)C, C,
+                       BN_NB, p0, p1);
+    hipLaunchKernelGGL(bn_fold_partials, dim3(C), dim3(BN_THREADS), 0,
+                       stream, p0, p1, BN_NB, C, sums, sumsqs);
   }
   hipLaunchKernelGGL(bn_fwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
                      stream, sums, sumsqs, M, C, (const float*)gamma,
@@ -402,6 +441,7 @@ extern "C" void launch_bn_bwd(
     long long M, int C, const void* gamma, const void* save_mean,
     const void* save_inv_std, int relu, void* dgamma, void* dbeta,
     void* workspace,  // 5*C floats: sum_dy, sum_dy_xhat, coef_a/b/d
+    void* partials,   // 2 * C * BN_NB floats
     hipStream_t stream) {
   float* ws = (float*)workspace;
   float* sum_dy = ws;
@@ -409,13 +449,15 @@ extern "C" void launch_bn_bwd(
   float* coef_a = ws + 2 * C;
   float* coef_b = ws + 3 * C;
   float* coef_d = ws + 4 * C;
-  (void)hipMemsetAsync(sum_dy, 0, 2 * C * sizeof(float), stream);
-  hipLaunchKernelGGL(bn_bwd_reduce,
-                     dim3(bn_elem_grid(M * (long long)C, C)),
-                     dim3(BN_THREADS), 0, stream, (const uint16_t*)dy,
-                     (const uint16_t*)x, (const uint16_t*)y,
-                     M * (long long)C, C, (const float*)save_mean,
-                     (const float*)save_inv_std, relu, sum_dy, sum_dy_xhat);
+  float* p0 = (float*)partials;
+  float* p1 = p0 + (long long)C * BN_NB;
+  hipLaunchKernelGGL(bn_bwd_reduce, dim3(BN_NB), dim3(BN_THREADS), 0,
+                     stream, (const uint16_t*)dy, (const uint16_t*)x,
+                     (const uint16_t*)y, M * (long long)C, C,
+                     (const float*)save_mean, (const float*)save_inv_std,
+                     relu, BN_NB, p0, p1);
+  hipLaunchKernelGGL(bn_fold_partials, dim3(C), dim3(BN_THREADS), 0,
+                     stream, p0, p1, BN_NB, C, sum_dy, sum_dy_xhat);
   hipLaunchKernelGGL(bn_bwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
                      stream, sum_dy, sum_dy_xhat, M, C, (const float*)gamma,
                      (const float*)save_inv_std, (float*)dgamma,
