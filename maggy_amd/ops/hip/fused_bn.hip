@@ -379,9 +379,23 @@ void bn_bwd_apply_vec8(const uint16_t* __restrict__ dy,
 
 // ------------------------------------------------- host launch wrappers
 
-// partial-row count for the two-stage reduction: fixed power of 2 so the
-// element stride NB*BN_THREADS*8 divides by every power-of-2 C <= 2048
+// Max partial rows for the two-stage reduction (allocation layout size).
+// The actual row count adapts: no more blocks than the sweep needs, and
+// capped so the partial buffer stays <= ~4 MB for large C (2*C*NB*4B).
+// Any power-of-2 NB keeps the fixed-channel stride contract because every
+// power-of-2 C <= 2048 divides BN_THREADS*8.
 #define BN_NB 1024
+
+static inline int bn_pick_nb(long long total, int C) {
+  long long nb = (total / 8 + BN_THREADS - 1) / BN_THREADS;
+  long long cap = (4LL << 20) / (8LL * C);  // partials <= 4 MB
+  if (nb > cap) nb = cap;
+  if (nb > BN_NB) nb = BN_NB;
+  if (nb < 64) nb = 64;
+  // round down to a multiple of 4 (fold kernel reads float4 rows)
+  nb &= ~3LL;
+  return (int)nb;
+}
 
 static inline long long bn_gcd(long long a, long long b) {
   while (b) { long long t = a % b; a = b; b = t; }
@@ -417,13 +431,14 @@ extern "C" void launch_bn_fwd(
   float* scale = ws + 4 * C;
   float* shift = ws + 5 * C;
   if (training) {
+    const int nb = bn_pick_nb(M * (long long)C, C);
     float* p0 = (float*)partials;
-    float* p1 = p0 + (long long)C * BN_NB;
-    hipLaunchKernelGGL(bn_sum_partial, dim3(BN_NB), dim3(BN_THREADS), 0,
+    float* p1 = p0 + (long long)C * BN_NB;  // fixed allocation layout
+    hipLaunchKernelGGL(bn_sum_partial, dim3(nb), dim3(BN_THREADS), 0,
                        stream, (const uint16_t*)x, M * (long long)C, C,
-                       BN_NB, p0, p1);
+                       nb, p0, p1);
     hipLaunchKernelGGL(bn_fold_partials, dim3(C), dim3(BN_THREADS), 0,
-                       stream, p0, p1, BN_NB, C, sums, sumsqs);
+                       stream, p0, p1, nb, C, sums, sumsqs);
   }
   hipLaunchKernelGGL(bn_fwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
                      stream, sums, sumsqs, M, C, (const float*)gamma,
@@ -449,15 +464,16 @@ extern "C" void launch_bn_bwd(
   float* coef_a = ws + 2 * C;
   float* coef_b = ws + 3 * C;
   float* coef_d = ws + 4 * C;
+  const int nb = bn_pick_nb(M * (long long)C, C);
   float* p0 = (float*)partials;
-  float* p1 = p0 + (long long)C * BN_NB;
-  hipLaunchKernelGGL(bn_bwd_reduce, dim3(BN_NB), dim3(BN_THREADS), 0,
+  float* p1 = p0 + (long long)C * BN_NB;  // fixed allocation layout
+  hipLaunchKernelGGL(bn_bwd_reduce, dim3(nb), dim3(BN_THREADS), 0,
                      stream, (const uint16_t*)dy, (const uint16_t*)x,
                      (const uint16_t*)y, M * (long long)C, C,
                      (const float*)save_mean, (const float*)save_inv_std,
-                     relu, BN_NB, p0, p1);
+                     relu, nb, p0, p1);
   hipLaunchKernelGGL(bn_fold_partials, dim3(C), dim3(BN_THREADS), 0,
-                     stream, p0, p1, BN_NB, C, sum_dy, sum_dy_xhat);
+                     stream, p0, p1, nb, C, sum_dy, sum_dy_xhat);
   hipLaunchKernelGGL(bn_bwd_finalize, dim3((C + 255) / 256), dim3(256), 0,
                      stream, sum_dy, sum_dy_xhat, M, C, (const float*)gamma,
                      (const float*)save_inv_std, (float*)dgamma,
