@@ -92,6 +92,11 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
                         default=json_default_numpy,
                     )
                 )
+            from maggy_amd import tensorboard
+
+            tensorboard._register(trial_dir)
+            if experiment_type == "optimization":
+                tensorboard._write_hparams(parameters, trial_id)
 
             start = time.time()
             early_stopped = False
