@@ -19,6 +19,7 @@ SOURCES = [
     os.path.join(HIP_DIR, "bindings.cpp"),
     os.path.join(HIP_DIR, "maggy_kernels.hip"),
     os.path.join(HIP_DIR, "fused_bn.hip"),
+    os.path.join(HIP_DIR, "fused_rms.hip"),
 ]
 
 

@@ -1,0 +1,101 @@
+"""Fused RMSNorm and SwiGLU modules on the HIP kernels (Llama hot path).
+
+The eager RMSNorm (x.float() -> pow -> mean -> rsqrt -> muls -> cast) costs
+~6 fp32 sweeps per call; the fused kernel is one bf16 read + write with
+fp32 accumulation (measured 1.7 ms -> ~0.12 ms f+b per [32768, 2048]
+call).  SwiGLU collapses silu(g)*u to one read-pair + write.
+
+Both fall back to the eager torch path off-GPU / non-bf16.
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from maggy_amd import ops
+
+RMS_NB = 1024  # must match RMS_NB in fused_rms.hip
+
+
+class _RMSNormFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = ops.require_ext()
+        shape = x.shape
+        D = shape[-1]
+        x2 = x.contiguous().view(-1, D)
+        R = x2.shape[0]
+        y = torch.empty_like(x2)
+        inv_rms = torch.empty(R, dtype=torch.float32, device=x.device)
+        ext.rms_fwd(x2, weight, y, inv_rms, R, D, eps)
+        ctx.save_for_backward(x2, weight, inv_rms)
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.require_ext()
+        x2, weight, inv_rms = ctx.saved_tensors
+        D = x2.shape[-1]
+        R = x2.shape[0]
+        dy2 = dy.contiguous().view(-1, D)
+        dx = torch.empty_like(dy2)
+        dw = torch.empty(D, dtype=torch.float32, device=dy.device)
+        partials = torch.empty(D * RMS_NB, dtype=torch.float32,
+                               device=dy.device)
+        ext.rms_bwd(dy2, x2, weight, inv_rms, dx, partials, dw, R, D)
+        return dx.view(dy.shape), dw, None
+
+
+class MaggyRMSNorm(nn.Module):
+    """Drop-in RMSNorm (weight fp32, same math as models.llama.RMSNorm)."""
+
+    def __init__(self, dim, eps=1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def _use_fused(self, x):
+        D = x.shape[-1]
+        return (x.is_cuda and x.dtype == torch.bfloat16
+                and D % 2048 == 0 and D <= 8192 and ops.has_ext())
+
+    def forward(self, x):
+        if self._use_fused(x):
+            # the kernel wants an fp32 gamma; .float() is differentiable so
+            # a bf16-cast module (model.to(bfloat16)) still gets its grad
+            w = self.weight if self.weight.dtype == torch.float32 \
+                else self.weight.float()
+            return _RMSNormFunction.apply(x, w, self.eps)
+        dt = x.dtype
+        xf = x.float()
+        xf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (xf * self.weight.float()).to(dt)
+
+
+class _SwiGLUFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g, u):
+        ext = ops.require_ext()
+        g = g.contiguous()
+        u = u.contiguous()
+        out = torch.empty_like(g)
+        ext.swiglu_fwd(g, u, out)
+        ctx.save_for_backward(g, u)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.require_ext()
+        g, u = ctx.saved_tensors
+        dy = dy.contiguous()
+        dg = torch.empty_like(g)
+        du = torch.empty_like(u)
+        ext.swiglu_bwd(dy, g, u, dg, du)
+        return dg, du
+
+
+def swiglu(g, u):
+    """silu(g) * u, fused on GPU bf16."""
+    if g.is_cuda and g.dtype == torch.bfloat16 and g.numel() % 8 == 0 \
+            and ops.has_ext():
+        return _SwiGLUFunction.apply(g, u)
+    return F.silu(g) * u

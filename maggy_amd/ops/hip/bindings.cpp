@@ -25,6 +25,15 @@ extern "C" void launch_bn_bwd(const void*, const void*, const void*, void*,
                               void*, long long, int, const void*,
                               const void*, const void*, int, void*, void*,
                               void*, void*, hipStream_t);
+extern "C" void launch_rms_fwd(const void*, const void*, void*, void*,
+                               long long, int, float, hipStream_t);
+extern "C" void launch_rms_bwd(const void*, const void*, const void*,
+                               const void*, void*, void*, void*, long long,
+                               int, hipStream_t);
+extern "C" void launch_swiglu_fwd(const void*, const void*, void*,
+                                  long long, hipStream_t);
+extern "C" void launch_swiglu_bwd(const void*, const void*, const void*,
+                                  void*, void*, long long, hipStream_t);
 extern "C" void launch_reduce_sum_f32(const float*, long long, float*,
                                       hipStream_t);
 extern "C" void launch_reduce_sum_bf16(const void*, long long, float*,
@@ -134,6 +143,45 @@ void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor y,
                 current_stream());
 }
 
+constexpr int64_t kRmsNB = 1024;  // must match RMS_NB in fused_rms.hip
+
+void rms_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor y,
+             torch::Tensor inv_rms, int64_t R, int64_t D, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  TORCH_CHECK(w.dtype() == torch::kFloat32);
+  TORCH_CHECK(D % 2048 == 0 && D / 2048 <= 4,
+              "fused RMSNorm requires D a multiple of 2048, <= 8192");
+  launch_rms_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                 inv_rms.data_ptr(), R, (int)D, (float)eps,
+                 current_stream());
+}
+
+void rms_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+             torch::Tensor inv_rms, torch::Tensor dx,
+             torch::Tensor dw_partials, torch::Tensor dw, int64_t R,
+             int64_t D) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16);
+  TORCH_CHECK(dw_partials.numel() >= D * kRmsNB);
+  launch_rms_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                 inv_rms.data_ptr(), dx.data_ptr(), dw_partials.data_ptr(),
+                 dw.data_ptr(), R, (int)D, current_stream());
+}
+
+void swiglu_fwd(torch::Tensor g, torch::Tensor u, torch::Tensor out) {
+  TORCH_CHECK(g.is_cuda() && g.dtype() == torch::kBFloat16);
+  TORCH_CHECK(g.numel() % 8 == 0);
+  launch_swiglu_fwd(g.data_ptr(), u.data_ptr(), out.data_ptr(), g.numel(),
+                    current_stream());
+}
+
+void swiglu_bwd(torch::Tensor dy, torch::Tensor g, torch::Tensor u,
+                torch::Tensor dg, torch::Tensor du) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16);
+  launch_swiglu_bwd(dy.data_ptr(), g.data_ptr(), u.data_ptr(),
+                    dg.data_ptr(), du.data_ptr(), dy.numel(),
+                    current_stream());
+}
+
 void reduce_sum(torch::Tensor in, torch::Tensor out) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous());
   TORCH_CHECK(out.is_cuda() && out.dtype() == torch::kFloat32);
@@ -169,6 +217,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused NHWC bf16 BatchNorm fwd (+residual +relu)");
   m.def("bn_bwd", &bn_bwd,
         "fused NHWC bf16 BatchNorm bwd (+relu mask +residual grad)");
+  m.def("rms_fwd", &rms_fwd, "fused bf16 RMSNorm forward");
+  m.def("rms_bwd", &rms_bwd, "fused bf16 RMSNorm backward");
+  m.def("swiglu_fwd", &swiglu_fwd, "fused silu(g)*u forward");
+  m.def("swiglu_bwd", &swiglu_bwd, "fused silu(g)*u backward");
   m.def("reduce_sum", &reduce_sum, "scalar sum reduction");
   m.def("reduce_max", &reduce_max, "scalar max reduction");
 }
