@@ -101,8 +101,8 @@ extern "C" __global__ __launch_bounds__(RMS_THREADS)
 void rms_bwd(const uint16_t* __restrict__ dy,
              const uint16_t* __restrict__ x, const float* __restrict__ w,
              const float* __restrict__ inv_rms, uint16_t* __restrict__ dx,
-             float* __restrict__ dw_partials,  // [D][RMS_NB]
-             long long R, int D) {
+             float* __restrict__ dw_partials,  // [D][NB]
+             long long R, int D, int NB) {
   const int K = D / (RMS_THREADS * 8);
   float wv[RMS_MAXK][8];
   float dwacc[RMS_MAXK][8];
@@ -169,8 +169,7 @@ void rms_bwd(const uint16_t* __restrict__ dy,
     const int base = (threadIdx.x + k * RMS_THREADS) * 8;
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      dw_partials[(long long)(base + j) * RMS_NB + blockIdx.x] =
-          dwacc[k][j];
+      dw_partials[(long long)(base + j) * NB + blockIdx.x] = dwacc[k][j];
     }
   }
 }
@@ -312,7 +311,7 @@ extern "C" void launch_rms_bwd(const void* dy, const void* x, const void* w,
   hipLaunchKernelGGL(rms_bwd, dim3(nb), dim3(RMS_THREADS), 0, stream,
                      (const uint16_t*)dy, (const uint16_t*)x,
                      (const float*)w, (const float*)inv_rms, (uint16_t*)dx,
-                     (float*)dw_partials, R, D);
+                     (float*)dw_partials, R, D, nb);
   hipLaunchKernelGGL(rms_fold, dim3(D), dim3(RMS_THREADS), 0, stream,
                      (float*)dw_partials, nb, D, (float*)dw);
 }
