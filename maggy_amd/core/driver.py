@@ -83,6 +83,7 @@ class OptimizationDriver:
         self.experiment_done = False
         self.result = {}
         self.maggy_log = ""
+        self.executor_logs = ""  # jupyter-style log stream (parity LOG msg)
         self.job_start = None
         self.job_end = None
         self.duration = None
@@ -228,6 +229,8 @@ class OptimizationDriver:
         self._final_store.append(trial)
         self._trial_store.pop(trial_id, None)
         w.trial_id = None
+        if logs:
+            self.executor_logs += logs
         self._update_result(trial)
         self.maggy_log = self.log_string()
         self.log(self.maggy_log)
@@ -353,6 +356,13 @@ class OptimizationDriver:
         self.result["avg"] = sum(numeric) / len(numeric) if numeric else None
         if trial.early_stop:
             self.result["early_stopped"] += 1
+
+    def get_logs(self):
+        """Drain the accumulated executor log stream plus a progress
+        snapshot (parity: LOG request, rpc.py:490-502)."""
+        logs = self.executor_logs
+        self.executor_logs = ""
+        return self.log_string(), logs
 
     def log_string(self):
         return (

@@ -100,8 +100,11 @@ class BaseAsyncBO(AbstractOptimizer):
                 new_trial = self.create_trial(
                     params, sample_type="warmup", run_budget=budget)
             else:
+                # BOHB-style: sample from the largest budget with a model
+                model_budget = max(
+                    (b for b in self.models if b <= budget), default=0)
                 new_trial = self._model_or_random(
-                    budget=0, run_budget=budget)
+                    budget=model_budget, run_budget=budget)
         else:
             parent = self._find_trial(parent_id)
             params = dict(parent.params)
@@ -132,10 +135,18 @@ class BaseAsyncBO(AbstractOptimizer):
         """Build the surrogate training matrix from finalized trials in the
         unit hypercube, min-convention y; busy (running) locations are
         appended with a constant-liar imputed metric (parity
-        base.py:400-457)."""
+        base.py:400-457).  With a pruner (multi-fidelity), ``budget``
+        selects that fidelity's observations (BOHB keeps one model per
+        budget); budget=0 or no matches uses every finalized trial."""
         X, y = [], []
         sign = -1.0 if self.direction == "max" else 1.0
-        for t in self.final_store:
+        pool = self.final_store
+        if budget:
+            at_budget = [t for t in self.final_store
+                         if t.params.get("budget") == budget]
+            if at_budget:
+                pool = at_budget
+        for t in pool:
             if t.final_metric is None:
                 continue
             params = {k: v for k, v in t.params.items() if k != "budget"}

@@ -122,6 +122,8 @@ class TorchDistributedTrainingDriver:
             "world_size": self.world_size,
             "duration_ms": util.seconds_to_milliseconds(
                 time.time() - self.job_start),
+            # rank-0 heartbeat metric history (step, value)
+            "metric_history": metric_stream.get(0, []),
         }
         Environment.get_instance().dump(
             json.dumps(self.result, default=json_default_numpy),
