@@ -125,9 +125,6 @@ def make_llama_step(args, device, world, size):
 
 
 def run_train(args):
-    # dmabuf IPC for RCCL; heuristic MIOpen find keeps warmup in seconds
-    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
     rank, world, local_rank = dist_env()
     dist = setup_dist(world, local_rank)
     device = torch.device("cuda", local_rank)
@@ -237,6 +234,9 @@ def run_asha(args):
 
 
 if __name__ == "__main__":
+    # dmabuf IPC for RCCL; heuristic MIOpen find keeps warmup in seconds
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
     args = parse_args()
     if args.mode == "asha":
         run_asha(args)
