@@ -9,7 +9,10 @@ def resnet_trial_fn(hparams, reporter):
     from maggy_amd.ops import FusedSGD
 
     device = torch.device("cuda:0")
-    torch.backends.cudnn.benchmark = True
+    # immediate-mode conv selection (MIOPEN_FIND_MODE=FAST heuristics):
+    # benchmark=True would run a ~2 min exhaustive MIOpen find inside the
+    # first trial, which belongs to the measured trials/hr
+    torch.backends.cudnn.benchmark = False
     model = resnet50().to(device, memory_format=torch.channels_last)
     opt = FusedSGD(model.parameters(), lr=hparams["lr"],
                    momentum=hparams.get("momentum", 0.9))
