@@ -67,9 +67,12 @@ class _FusedOptimizerBase(torch.optim.Optimizer):
         return out
 
     def _build_tables(self, pairs):
-        """Group params by (grad dtype, has bf16 mirror) and pack the device
-        tables.  Pointer stability: rebuilt whenever any data_ptr changes."""
+        """Group params by (param_group, grad dtype, has bf16 mirror) and
+        pack the device tables — one kernel launch per (group, mode) so
+        per-group hyperparameters apply.  Pointer stability: rebuilt
+        whenever any data_ptr changes."""
         device = pairs[0][1].device
+        group_index = {id(g): i for i, g in enumerate(self.param_groups)}
         modes = {}
         for group, p in pairs:
             st = self.state[p]
@@ -78,7 +81,7 @@ class _FusedOptimizerBase(torch.optim.Optimizer):
                 st = self.state[p]
             grad_bf16 = p.grad.dtype == torch.bfloat16
             bf16_param = "master" in st
-            key = (grad_bf16, bf16_param)
+            key = (group_index[id(group)], grad_bf16, bf16_param)
             master = st["master"] if bf16_param else p
             meta = (
                 master.data_ptr(),
@@ -90,9 +93,10 @@ class _FusedOptimizerBase(torch.optim.Optimizer):
             )
             modes.setdefault(key, []).append(meta)
         tables = []
-        for (grad_bf16, bf16_param), metas in modes.items():
+        for (gi, grad_bf16, bf16_param), metas in sorted(modes.items()):
             t, c, n = _pack_tables(metas, device)
             tables.append({
+                "group": self.param_groups[gi],
                 "grad_bf16": grad_bf16,
                 "tensors": t, "chunks": c, "n_chunks": n,
             })
@@ -141,11 +145,6 @@ class FusedAdam(_FusedOptimizerBase):
         if not pairs:
             return loss
         self._step_count += 1
-        g0 = self.param_groups[0]
-        lr, (b1, b2), eps, wd = (g0["lr"], g0["betas"], g0["eps"],
-                                 g0["weight_decay"])
-        bc1 = 1.0 - b1 ** self._step_count
-        bc2 = 1.0 - b2 ** self._step_count
         if pairs[0][1].is_cuda:
             ext = ops.require_ext()
             tables = self._ensure_tables(pairs)
@@ -153,15 +152,19 @@ class FusedAdam(_FusedOptimizerBase):
             if self.max_grad_norm:
                 norm = self._grad_norm_launch(ext, tables)
             for tb in tables:
+                g = tb["group"]
+                b1, b2 = g["betas"]
                 ext.multi_fused_adam(
                     tb["chunks"], tb["n_chunks"], tb["tensors"],
-                    tb["grad_bf16"], lr, b1, b2, eps, wd, bc1, bc2, norm,
+                    tb["grad_bf16"], g["lr"], b1, b2, g["eps"],
+                    g["weight_decay"], 1.0 - b1 ** self._step_count,
+                    1.0 - b2 ** self._step_count, norm,
                     self.max_grad_norm or 0.0, 1.0 / self.grad_scale)
         else:
-            self._cpu_step(pairs, lr, b1, b2, eps, wd, bc1, bc2)
+            self._cpu_step(pairs)
         return loss
 
-    def _cpu_step(self, pairs, lr, b1, b2, eps, wd, bc1, bc2):
+    def _cpu_step(self, pairs):
         """Same math in torch (CPU fallback; also the GPU test reference)."""
         clip = 1.0
         if self.max_grad_norm:
@@ -172,7 +175,11 @@ class FusedAdam(_FusedOptimizerBase):
             if nrm > self.max_grad_norm:
                 clip = self.max_grad_norm / (nrm + 1e-6)
         gscale = clip / self.grad_scale
-        for _, p in pairs:
+        for group, p in pairs:
+            lr, (b1, b2), eps, wd = (group["lr"], group["betas"],
+                                     group["eps"], group["weight_decay"])
+            bc1 = 1.0 - b1 ** self._step_count
+            bc2 = 1.0 - b2 ** self._step_count
             st = self.state[p]
             if not st:
                 self._init_param_state(p)
@@ -213,10 +220,6 @@ class FusedSGD(_FusedOptimizerBase):
         if not pairs:
             return loss
         self._step_count += 1
-        g0 = self.param_groups[0]
-        lr, mom, wd, damp, nesterov = (
-            g0["lr"], g0["momentum"], g0["weight_decay"], g0["dampening"],
-            g0["nesterov"])
         first = self._step_count == 1
         if pairs[0][1].is_cuda:
             ext = ops.require_ext()
@@ -225,15 +228,18 @@ class FusedSGD(_FusedOptimizerBase):
             if self.max_grad_norm:
                 norm = self._grad_norm_launch(ext, tables)
             for tb in tables:
+                g = tb["group"]
                 ext.multi_fused_sgd(
                     tb["chunks"], tb["n_chunks"], tb["tensors"],
-                    tb["grad_bf16"], lr, mom, wd, damp, nesterov, first,
-                    norm, self.max_grad_norm or 0.0, 1.0 / self.grad_scale)
+                    tb["grad_bf16"], g["lr"], g["momentum"],
+                    g["weight_decay"], g["dampening"], g["nesterov"],
+                    first, norm, self.max_grad_norm or 0.0,
+                    1.0 / self.grad_scale)
         else:
-            self._cpu_step(pairs, lr, mom, wd, damp, nesterov, first)
+            self._cpu_step(pairs, first)
         return loss
 
-    def _cpu_step(self, pairs, lr, mom, wd, damp, nesterov, first):
+    def _cpu_step(self, pairs, first):
         clip = 1.0
         if self.max_grad_norm:
             total = 0.0
@@ -243,7 +249,10 @@ class FusedSGD(_FusedOptimizerBase):
             if nrm > self.max_grad_norm:
                 clip = self.max_grad_norm / (nrm + 1e-6)
         gscale = clip / self.grad_scale
-        for _, p in pairs:
+        for group, p in pairs:
+            lr, mom, wd, damp, nesterov = (
+                group["lr"], group["momentum"], group["weight_decay"],
+                group["dampening"], group["nesterov"])
             st = self.state[p]
             if not st:
                 self._init_param_state(p)
