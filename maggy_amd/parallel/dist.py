@@ -39,9 +39,15 @@ def init_process_group(backend=None, timeout_s=120):
 
 
 def wrap_ddp(module, bucket_cap_mb=None, device_ids=None,
-             find_unused_parameters=False):
+             find_unused_parameters=False, bf16_allreduce=None):
     """DDP wrap with xGMI-tuned bucketing (bucketed all-reduce overlapped
-    with backward — reference call-site N2)."""
+    with backward — reference call-site N2).
+
+    ``bf16_allreduce`` compresses fp32 gradient buckets to bf16 for the
+    all-reduce (half the per-link xGMI bytes).  Default: on when the model
+    params are fp32 (autocast training) and a GPU process group is up;
+    bf16-param models already reduce in bf16.
+    """
     if bucket_cap_mb is None:
         bucket_cap_mb = XGMI_BUCKET_CAP_MB
     kwargs = dict(
@@ -49,10 +55,19 @@ def wrap_ddp(module, bucket_cap_mb=None, device_ids=None,
         gradient_as_bucket_view=True,
         find_unused_parameters=find_unused_parameters,
     )
-    if torch.cuda.is_available() and next(
-            module.parameters()).is_cuda:
+    first_param = next(module.parameters())
+    on_gpu = torch.cuda.is_available() and first_param.is_cuda
+    if on_gpu:
         kwargs["device_ids"] = device_ids or [torch.cuda.current_device()]
-    return torch.nn.parallel.DistributedDataParallel(module, **kwargs)
+    ddp = torch.nn.parallel.DistributedDataParallel(module, **kwargs)
+    if bf16_allreduce is None:
+        bf16_allreduce = on_gpu and first_param.dtype == torch.float32
+    if bf16_allreduce:
+        from torch.distributed.algorithms.ddp_comm_hooks.default_hooks \
+            import bf16_compress_hook
+
+        ddp.register_comm_hook(state=None, hook=bf16_compress_hook)
+    return ddp
 
 
 def barrier():
