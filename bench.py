@@ -62,7 +62,7 @@ def make_resnet_step(args, device, world):
     from maggy_amd.models import resnet50
     from maggy_amd.ops import FusedAdam
 
-    batch = args.batch or 256
+    batch = args.batch or 512
     model = resnet50().to(device, memory_format=torch.channels_last)
     if world > 1:
         from maggy_amd.parallel.dist import wrap_ddp
@@ -93,7 +93,7 @@ def make_llama_step(args, device, world, size):
 
     cfg_model = (LlamaConfig.llama3_8b() if size == "8b"
                  else LlamaConfig.small_1b())
-    batch = args.batch or (2 if size == "8b" else 8)
+    batch = args.batch or (4 if size == "8b" else 8)
     seq = args.seq_len
     # construct directly on the GPU (init kernels run on-device; 8B fp32
     # transient fits easily in 288 GB HBM3E), then cast params to bf16
