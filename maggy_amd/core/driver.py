@@ -88,6 +88,9 @@ class OptimizationDriver:
         self.job_end = None
         self.duration = None
         self._idle_workers = {}  # worker_id -> idle_start time
+        # scheduling telemetry: first trial assignment / last finalization
+        self.first_assign_ts = None
+        self.last_final_ts = None
 
     # ------------------------------------------------------------------
     @staticmethod
@@ -190,6 +193,8 @@ class OptimizationDriver:
             self._idle_workers[w.worker_id] = time.time()
         else:
             trial.start = time.time()
+            if self.first_assign_ts is None:
+                self.first_assign_ts = trial.start
             trial.status = Trial.SCHEDULED
             self.add_trial(trial)
             self.pool.assign(w, trial)
@@ -229,6 +234,7 @@ class OptimizationDriver:
         self._final_store.append(trial)
         self._trial_store.pop(trial_id, None)
         w.trial_id = None
+        self.last_final_ts = time.time()
         if logs:
             self.executor_logs += logs
         self._update_result(trial)

@@ -121,6 +121,7 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
                 retval = util.handle_return_val(
                     retval, trial_dir, optimization_key, trial_log_file
                 )
+                reporter.log("Finished Trial: {}".format(trial_id), False)
                 conn.send((M.FINAL, worker_id, trial_id, retval,
                            time.time() - start, early_stopped, reporter.logs))
             except EarlyStopException as e:

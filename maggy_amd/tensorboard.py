@@ -15,13 +15,22 @@ import threading
 _lock = threading.Lock()
 _logdir = None
 _writer = None
+_writer_cls = "unprobed"  # probe once per process: ~2 s on a miss
 
 
 def _try_writer(logdir):
-    try:
-        from torch.utils.tensorboard import SummaryWriter
+    global _writer_cls
+    if _writer_cls == "unprobed":
+        try:
+            from torch.utils.tensorboard import SummaryWriter
 
-        return SummaryWriter(log_dir=logdir)
+            _writer_cls = SummaryWriter
+        except Exception:
+            _writer_cls = None
+    if _writer_cls is None:
+        return None
+    try:
+        return _writer_cls(log_dir=logdir)
     except Exception:
         return None
 
