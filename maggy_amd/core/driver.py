@@ -121,10 +121,36 @@ class OptimizationDriver:
         return self._trial_store.get(trial_id)
 
     # ------------------------------------------------------------------
+    def resume_from(self, exp_dir):
+        """Preload finalized trials from a previous experiment directory
+        (the reference left this latent: Trial.from_json existed but was
+        never called).  Call before run_experiment; the
+        controller is told via ``on_resume`` so it does not re-run the
+        completed configurations."""
+        from maggy_amd.utils.checkpoint import load_finished_trials
+
+        loaded = [t for t in load_finished_trials(exp_dir)
+                  if t is not None and t.status == Trial.FINALIZED]
+        for t in loaded:
+            self._final_store.append(t)
+            self._update_result(t)
+        self._resumed = len(loaded)
+        self.log("resumed {} finalized trials from {}".format(
+            len(loaded), exp_dir))
+        return len(loaded)
+
     def run_experiment(self, train_fn, payload_extra=None):
         """Run the experiment to completion; returns the result dict."""
         self.job_start = time.time()
         self.controller._initialize(exp_dir=self.log_dir)
+        resumed = getattr(self, "_resumed", 0)
+        if resumed:
+            if hasattr(self.controller, "on_resume"):
+                self.controller.on_resume(self._final_store)
+            else:
+                raise NotImplementedError(
+                    "{} does not support resume".format(
+                        self.controller.name()))
         payload = {
             "train_fn": train_fn,
             "model": getattr(self.config, "model", None),

@@ -127,6 +127,19 @@ class BaseAsyncBO(AbstractOptimizer):
     def _budget(trial):
         return int(trial.params.get("budget", 0)) if trial else 0
 
+    def on_resume(self, finalized):
+        """Experiment resume: the surrogate trains on the preloaded
+        final_store; shrink the warmup buffer accordingly and rebuild the
+        model."""
+        if self.pruner is not None:
+            raise NotImplementedError(
+                "resume with a pruner is not supported")
+        del self.warmup_buffer[:len(finalized)]
+        try:
+            self.update_model(budget=0)
+        except Exception as e:
+            self._log("resume update_model failed: {}".format(e))
+
     def finalize_experiment(self, trials):
         return
 

@@ -61,5 +61,13 @@ class RandomSearch(AbstractOptimizer):
             return self.trial_store[trial_id]
         raise KeyError("Trial {} not found".format(trial_id))
 
+    def on_resume(self, finalized):
+        """Experiment resume: skip as many pre-sampled configs as have
+        already finalized (random search has no state beyond the buffer)."""
+        if self.pruner is not None:
+            raise NotImplementedError(
+                "resume with a pruner is not supported")
+        del self.config_buffer[:len(finalized)]
+
     def finalize_experiment(self, trials):
         return
