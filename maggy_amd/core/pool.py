@@ -142,7 +142,8 @@ class TrialPool:
     def assign(self, w, trial):
         w.trial_id = trial.trial_id
         w.ring.clear_stop()
-        w.conn.send((M.TRIAL, trial.trial_id, trial.params))
+        w.conn.send((M.TRIAL, trial.trial_id, trial.params,
+                     dict(trial.info_dict)))
 
     def request_stop(self, trial_id):
         """Flag a running trial for early stop via its worker's stop word."""

@@ -68,7 +68,7 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
                 break
             if msg[0] != M.TRIAL:
                 continue
-            _, trial_id, parameters = msg
+            _, trial_id, parameters, trial_info = msg
             parameters = dict(parameters)
             ablation_params = None
             if experiment_type == "ablation":
@@ -104,7 +104,13 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
             try:
                 reporter.log("Starting Trial: {}".format(trial_id), False)
                 reporter.log("Trial Configuration: {}".format(parameters), False)
-                extra = {}
+                extra = {
+                    "trial_dir": trial_dir,
+                    # promoted trials (ASHA/Hyperband) can continue from
+                    # the parent's checkpoint instead of restarting
+                    "parent_checkpoint": _parent_checkpoint(
+                        log_dir, trial_info),
+                }
                 if experiment_type == "ablation":
                     gen_extra = _ablation_generators(payload, parameters,
                                                      ablation_params)
@@ -142,6 +148,15 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
         reporter.close_logger()
         ring.close()
         conn.close()
+
+
+def _parent_checkpoint(log_dir, trial_info):
+    """Path to the promoted trial's parent checkpoint.pt, or None."""
+    parent = (trial_info or {}).get("parent_trial_id")
+    if not parent:
+        return None
+    path = os.path.join(log_dir, parent, "checkpoint.pt")
+    return path if os.path.exists(path) else None
 
 
 def _ablation_generators(payload, parameters, ablation_params):

@@ -88,10 +88,13 @@ class AbstractOptimizer(ABC):
 
     # -- helpers shared by concrete optimizers ---------------------------
     def create_trial(self, hparams, sample_type="random", run_budget=0,
-                     model_budget=None):
+                     model_budget=None, parent_trial_id=None):
         """Build a Trial, stamping scheduling metadata and injecting the
         training budget into the hparams when budget-based (parity:
-        abstractoptimizer.py:317-376)."""
+        abstractoptimizer.py:317-376).  ``parent_trial_id`` marks promoted
+        trials so the worker can hand the parent's checkpoint to the
+        training function (continuation — the reference restarts promoted
+        configs from scratch)."""
         info_dict = {
             "run_budget": run_budget,
             "sample_type": sample_type,
@@ -99,6 +102,8 @@ class AbstractOptimizer(ABC):
         }
         if model_budget is not None:
             info_dict["model_budget"] = model_budget
+        if parent_trial_id is not None:
+            info_dict["parent_trial_id"] = parent_trial_id
         if run_budget > 0:
             hparams = dict(hparams)
             hparams["budget"] = run_budget

@@ -72,7 +72,8 @@ class Asha(AbstractOptimizer):
                     self.reduction_factor ** new_rung)
                 params.pop("budget", None)
                 promote_trial = self.create_trial(
-                    params, sample_type="promoted", run_budget=budget)
+                    params, sample_type="promoted", run_budget=budget,
+                    parent_trial_id=old_trial.trial_id)
                 self.rungs.setdefault(new_rung, []).append(promote_trial)
                 self.promoted.setdefault(k, []).append(old_trial.trial_id)
                 self._log("promote {} -> rung {} (budget {})".format(

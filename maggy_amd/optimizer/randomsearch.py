@@ -48,7 +48,8 @@ class RandomSearch(AbstractOptimizer):
             params = dict(parent.params)
             params.pop("budget", None)
             new_trial = self.create_trial(
-                params, sample_type="promoted", run_budget=budget)
+                params, sample_type="promoted", run_budget=budget,
+                parent_trial_id=parent_id)
         self.pruner.report_trial(
             original_trial_id=parent_id, new_trial_id=new_trial.trial_id)
         return new_trial

@@ -66,3 +66,27 @@ def single_run_fn(model, dataset, hparams, reporter):
     reporter.broadcast(1.0, 0)
     reporter.broadcast(2.0, 1)
     return {"Metric": 3.0, "extra": 7}
+
+
+def continuation_fn(hparams, reporter, trial_dir, parent_checkpoint):
+    """Budget-trial that continues from the parent's checkpoint: the
+    'model' is a cumulative step counter persisted as checkpoint.pt."""
+    import torch
+
+    steps_done = 0
+    if parent_checkpoint is not None:
+        steps_done = torch.load(parent_checkpoint,
+                                weights_only=False)["model"]["steps"]
+    budget = int(hparams.get("budget", 1))
+    for s in range(budget):
+        steps_done += 1
+        reporter.broadcast(float(steps_done), s)
+
+    class _M:
+        def state_dict(self):
+            return {"steps": steps_done}
+
+    from maggy_amd.utils.checkpoint import save_checkpoint
+
+    save_checkpoint(trial_dir, _M(), step=steps_done)
+    return float(steps_done)
