@@ -326,6 +326,15 @@ class OptimizationDriver:
                 self.log("worker {} died {} times; not respawning".format(
                     w.worker_id, w.respawns))
                 w.process = None
+                # the held trial can never finish: mark it ERROR so
+                # budget-based controllers (ASHA rungs) don't wait forever
+                if lost_trial_id is not None:
+                    trial = self.get_trial(lost_trial_id)
+                    if trial is not None:
+                        with trial.lock:
+                            trial.status = Trial.ERROR
+                        self._error_store.append(trial)
+                        self._trial_store.pop(lost_trial_id, None)
                 w.trial_id = None
                 continue
             self.pool.respawn(w)

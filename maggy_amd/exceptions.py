@@ -91,12 +91,12 @@ class BroadcastStepValueError(Exception):
 class WorkerCrashError(Exception):
     """A trial-pool worker process died (nonzero exit / signal)."""
 
-    def __init__(self, worker_id, exitcode):
+    def __init__(self, worker_id, detail):
         super().__init__(
-            "Trial worker {} died with exit code {}.".format(worker_id, exitcode)
+            "Trial worker(s) {} died: {}".format(worker_id, detail)
         )
         self.worker_id = worker_id
-        self.exitcode = exitcode
+        self.detail = detail
 
 
 class HipExtensionMissingError(ImportError):
