@@ -92,3 +92,13 @@ def dist_zero_fn(module, hparams, reporter):
             assert torch.allclose(g, p.detach(), atol=1e-6), \
                 "zero shards diverged"
     return last
+
+
+def dist_crashing_fn(module, hparams, reporter):
+    """Rank 1 raises to exercise distributed failure propagation."""
+    import torch.distributed as dist
+
+    if dist.get_rank() == 1:
+        raise RuntimeError("deliberate rank failure")
+    reporter.broadcast(1.0, 0)
+    return 1.0

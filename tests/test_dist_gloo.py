@@ -31,3 +31,13 @@ def test_zero_sharded_optimizer_gloo(exp_dir):
         module=fns.TinyNet, num_gpus=2, zero_lvl=1, name="zero-gloo")
     res = experiment.lagom(fns.dist_zero_fn, cfg)
     assert res["final_metric_avg"] is not None
+
+
+@pytest.mark.timeout(180)
+def test_dist_rank_failure_raises(exp_dir):
+    from maggy_amd.exceptions import WorkerCrashError
+
+    cfg = TorchDistributedConfig(
+        module=fns.TinyNet, num_gpus=2, name="crash-gloo")
+    with pytest.raises(WorkerCrashError):
+        experiment.lagom(fns.dist_crashing_fn, cfg)

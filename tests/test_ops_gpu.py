@@ -167,5 +167,5 @@ def test_fused_adam_trains_a_model(ext):
             loss = torch.nn.functional.cross_entropy(model(x), y)
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.5, losses
