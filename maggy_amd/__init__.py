@@ -13,6 +13,7 @@ shims.
 """
 from maggy_amd.searchspace import Searchspace  # noqa: F401
 from maggy_amd.trial import Trial  # noqa: F401
+from maggy_amd import experiment  # noqa: F401  (lagom entry point)
 
 __version__ = "0.1.0"
 __all__ = ["Searchspace", "Trial", "experiment"]
