@@ -320,6 +320,146 @@ void gemm_tn_bf16_v2(const uint16_t* __restrict__ A,
   }
 }
 
+// ------------------------------------------------- v3: 1x8 wave split,
+// progressive consumption, glds flight ACROSS the tile flip.
+// Wave wn owns C[:, 32*wn..+32) (16 m-frags x 2 n-frags, 128 acc regs).
+// Per tile: phase 0 reads the wave's W slice (4 ds) + A rows [0,64)
+// (8 ds = the 12-read phase); phases 1-3 read A quarters q (8 ds each).
+// Staging order of tile t+1 during tile t: {W.h0, W.h1, A.h0, A.h1}
+// (one half-tile = 2 glds/wave per phase) matches consumption order, so
+// each tile's A.h1 stays in flight across the flip.  Per-wave FIFO count
+// (steady state, leftover = own A.h1 = 2 glds at tile entry):
+//   phase 1 end: outstanding 6, vmcnt(4) drains the CURRENT tile's A.h1
+//                just before phase 2 reads it
+//   phase 3 end: outstanding 8, vmcnt(2) drains W.h0,W.h1,A.h0 of t+1,
+//                leaving its A.h1 in flight (the invariant)
+// Raw barriers everywhere; one __shared__ array; no global loads in-loop.
+extern "C" __global__ __launch_bounds__(THREADS, 2)
+void gemm_tn_bf16_v3(const uint16_t* __restrict__ A,
+                     const uint16_t* __restrict__ W,
+                     uint16_t* __restrict__ C, int M, int N, int K) {
+  __shared__ uint16_t lds[2 * 2 * BM * BK];
+  const int nwgM = M / BM, nwgN = N / BN;
+  int wg = xcd_remap(blockIdx.x, nwgM * nwgN);
+  const int bm = (wg / nwgN) * BM;
+  const int bn = (wg % nwgN) * BN;
+  const int l = threadIdx.x;
+  const int wave = l >> 6;
+  const int lane = l & 63;
+  const int wn = wave;  // 1M x 8N
+
+  f32x4 acc[16][2];
+  #pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    acc[i][0] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    acc[i][1] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int KT = K / BK;
+
+  // stage one HALF-tile (rows [half*128, +128)) of one operand: each
+  // wave 2 KiB as 2 glds
+  auto stage_half = [&](uint16_t* ldsbase, const uint16_t* G,
+                        int grow_base, long long kbase, int half) {
+    #pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      const uint32_t s = (uint32_t)half * 16384 + (uint32_t)wave * 2048
+                       + (uint32_t)g * 1024 + (uint32_t)lane * 16;
+      const uint32_t o = swz(s);
+      const uint16_t* gp = G + ((long long)(grow_base + (o >> 7))) * K
+                         + kbase + ((o & 127) >> 1);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)gp,
+          (__attribute__((address_space(3))) uint32_t*)
+              (ldsbase + (half * 16384 + wave * 2048 + g * 1024) / 2),
+          16, 0, 0);
+    }
+  };
+
+  // prologue: tile 0 complete
+  #pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    stage_half(LDSW(0), W, bn, 0, h);
+    stage_half(LDSA(0), A, bm, 0, h);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  bf16x8 wfrag[2][2];
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    const uint16_t* ldsA = LDSA(cur);
+    const uint16_t* ldsW = LDSW(cur);
+    uint16_t* nA = LDSA(cur ^ 1);
+    uint16_t* nW = LDSW(cur ^ 1);
+    const long long knext = (long long)(kt + 1) * BK;
+    const bool staging = (kt + 1 < KT);
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      bf16x8 afrag[4][2];
+      if (q == 0) {
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          #pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            const int col = wn * 32 + nf * 16 + (lane & 15);
+            const uint32_t off = swz((uint32_t)col * 128 + kk * 64
+                                     + (lane >> 4) * 16);
+            wfrag[nf][kk] = *(const bf16x8*)((const char*)ldsW + off);
+          }
+      }
+      #pragma unroll
+      for (int a = 0; a < 4; ++a)
+        #pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int row = q * 64 + a * 16 + (lane & 15);
+          const uint32_t off = swz((uint32_t)row * 128 + kk * 64
+                                   + (lane >> 4) * 16);
+          afrag[a][kk] = *(const bf16x8*)((const char*)ldsA + off);
+        }
+      if (staging) {
+        // q: 0 -> W.h0, 1 -> W.h1, 2 -> A.h0, 3 -> A.h1
+        if (q < 2) stage_half(nW, W, bn, knext, q);
+        else       stage_half(nA, A, bm, knext, q - 2);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int a = 0; a < 4; ++a)
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          #pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[q * 4 + a][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[a][kk], wfrag[nf][kk], acc[q * 4 + a][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (q == 1) {
+        if (staging)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      } else if (q == 3) {
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  #pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    #pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      const int col = bn + wn * 32 + nf * 16 + (lane & 15);
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = bm + i * 16 + 4 * (lane >> 4) + r;
+        C[(long long)row * N + col] = gf2bf(acc[i][nf][r]);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------- host
 
 #define HIP_CHECK(x) do { hipError_t e = (x); if (e) { \
@@ -449,12 +589,16 @@ int main() {
     if (!refcheck(gemm_tn_bf16_v2, "v2", 512, 512, 512)) return 4;
     if (!refcheck(gemm_tn_bf16_v2, "v2", 1024, 512, 2048)) return 4;
     if (!refcheck(gemm_tn_bf16_v2, "v2", 1024, 1024, 4096)) return 4;
+    if (!refcheck(gemm_tn_bf16_v3, "v3", 512, 512, 512)) return 5;
+    if (!refcheck(gemm_tn_bf16_v3, "v3", 1024, 512, 2048)) return 5;
+    if (!refcheck(gemm_tn_bf16_v3, "v3", 1024, 1024, 4096)) return 5;
   }
   perf(gemm_tn_bf16, "v1", 4096, 4096, 4096, 20);
   perf(gemm_tn_bf16_v2, "v2", 4096, 4096, 4096, 20);
+  perf(gemm_tn_bf16_v3, "v3", 4096, 4096, 4096, 20);
   perf(gemm_tn_bf16, "v1", 16384, 8192, 2048, 10);
-  perf(gemm_tn_bf16_v2, "v2", 16384, 8192, 2048, 10);
+  perf(gemm_tn_bf16_v3, "v3", 16384, 8192, 2048, 10);
   perf(gemm_tn_bf16, "v1", 16384, 2048, 8192, 10);
-  perf(gemm_tn_bf16_v2, "v2", 16384, 2048, 8192, 10);
+  perf(gemm_tn_bf16_v3, "v3", 16384, 2048, 8192, 10);
   return 0;
 }
