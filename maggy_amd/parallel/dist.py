@@ -54,6 +54,10 @@ def wrap_ddp(module, bucket_cap_mb=None, device_ids=None,
         bucket_cap_mb=bucket_cap_mb,
         gradient_as_bucket_view=True,
         find_unused_parameters=find_unused_parameters,
+        # do not re-broadcast BN running stats and other buffers every
+        # forward (rank-local stats are the DDP norm; avoids a per-step
+        # broadcast of every buffer from rank 0)
+        broadcast_buffers=False,
     )
     first_param = next(module.parameters())
     on_gpu = torch.cuda.is_available() and first_param.is_cuda
