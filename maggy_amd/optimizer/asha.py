@@ -46,10 +46,12 @@ class Asha(AbstractOptimizer):
                     self.reduction_factor ** (self.max_rung + 1)))
 
     def get_suggestion(self, trial=None):
+        # stopping criterion: a trial reached the max rung (checked for
+        # every request so a resumed-complete experiment terminates
+        # without sampling fresh base-rung trials)
+        if self.rungs.get(self.max_rung):
+            return None
         if trial is not None:
-            # stopping criterion: a trial reached the max rung
-            if self.max_rung in self.rungs:
-                return None
             # try to promote, scanning rungs from high to low
             for k in range(self.max_rung - 1, -1, -1):
                 if k not in self.rungs:
@@ -95,6 +97,20 @@ class Asha(AbstractOptimizer):
         finished.sort(
             key=lambda t: t.final_metric, reverse=(self.direction == "max"))
         return finished[:number]
+
+    def on_resume(self, finalized):
+        """Rebuild the rung state from persisted trials: the rung index is
+        log_eta(budget/resource_min) (stamped into hparams by
+        create_trial), and a trial whose info_dict carries
+        parent_trial_id was created by promoting that parent."""
+        for t in finalized:
+            budget = int(t.params.get("budget", self.resource_min))
+            rung = int(round(math.log(
+                budget / self.resource_min, self.reduction_factor)))
+            self.rungs.setdefault(rung, []).append(t)
+            parent = (t.info_dict or {}).get("parent_trial_id")
+            if parent is not None and rung > 0:
+                self.promoted.setdefault(rung - 1, []).append(parent)
 
     def finalize_experiment(self, trials):
         return

@@ -90,3 +90,10 @@ def continuation_fn(hparams, reporter, trial_dir, parent_checkpoint):
 
     save_checkpoint(trial_dir, _M(), step=steps_done)
     return float(steps_done)
+
+
+def noisy_quadratic_fn(hparams, reporter):
+    """Deterministic objective for pool BO tests: peak at lr=0.06."""
+    v = 1.0 - (hparams["lr"] - 0.06) ** 2 * 100.0
+    reporter.broadcast(v, 0)
+    return v

@@ -43,12 +43,14 @@ def test_resume_unsupported_controller(exp_dir):
     experiment.lagom(fns.metric_eq_lr, cfg)
     run1, app = _run_dir(exp_dir)
 
+    sp2 = Searchspace(lr=("DISCRETE", [0.01, 0.1]))
     cfg2 = HyperparameterOptConfig(
-        num_trials=16, optimizer="asha", searchspace=sp,
+        num_trials=2, optimizer="gridsearch", searchspace=sp2,
         es_policy="none", num_workers=1, name="p2")
     d = OptimizationDriver(cfg2, app_id=app)
     d.resume_from(run1)
     import pytest
 
+    # GridSearch has no on_resume -> refuse rather than re-run configs
     with pytest.raises(NotImplementedError):
         d.run_experiment(fns.metric_eq_lr)
