@@ -18,6 +18,7 @@ from maggy_amd.trial import Trial
 
 class BaseAsyncBO(AbstractOptimizer):
     def __init__(self, num_warmup_trials=15, random_fraction=0.33,
+                 interim_results=False, interim_results_interval=3,
                  pruner=None, pruner_kwargs=None):
         super().__init__(pruner=pruner, pruner_kwargs=pruner_kwargs)
         self.num_warmup_trials = num_warmup_trials
@@ -26,6 +27,11 @@ class BaseAsyncBO(AbstractOptimizer):
         self.models = {}          # budget -> surrogate
         self.sampling_time = []
         self.imputed_metric = "cl_min"  # constant liar default
+        # interim-results augmentation (parity: reference base.py z=[x, n],
+        # :459-641): train the surrogate on heartbeat metrics at
+        # intermediate progress, each observation as [x..., progress]
+        self.interim_results = interim_results
+        self.interim_results_interval = interim_results_interval
 
     # -- subclass contract ---------------------------------------------
     def init_model(self, budget=0):
@@ -170,10 +176,28 @@ class BaseAsyncBO(AbstractOptimizer):
             X.append(x)
             y.append(sign * t.final_metric)
         n_fin = len(y)
+        if self.interim_results:
+            # augment each observation with a progress coordinate: final
+            # metrics at 1.0, heartbeat metrics at their fractional step
+            for i in range(n_fin):
+                X[i] = X[i] + [1.0]
+            for t in pool:
+                if t.final_metric is None or not t.metric_history:
+                    continue
+                hist = t.metric_history
+                T = len(hist)
+                params = {k: v for k, v in t.params.items()
+                          if k != "budget"}
+                xb = self.searchspace.transform(
+                    self.searchspace.dict_to_list(params),
+                    normalize_categorical=True)
+                for s in range(0, T - 1, self.interim_results_interval):
+                    X.append(xb + [(s + 1) / T])
+                    y.append(sign * hist[s])
         if include_busy and n_fin:
             liar = {
-                "cl_min": min(y), "cl_max": max(y),
-                "cl_mean": sum(y) / len(y),
+                "cl_min": min(y[:n_fin]), "cl_max": max(y[:n_fin]),
+                "cl_mean": sum(y[:n_fin]) / n_fin,
             }[self.imputed_metric]
             for t in self.trial_store.values():
                 params = {k: v for k, v in t.params.items()
@@ -184,6 +208,8 @@ class BaseAsyncBO(AbstractOptimizer):
                         normalize_categorical=True)
                 except (KeyError, ValueError):
                     continue
+                if self.interim_results:
+                    x = x + [1.0]
                 X.append(x)
                 y.append(liar)
         return np.asarray(X, dtype=float), np.asarray(y, dtype=float), n_fin

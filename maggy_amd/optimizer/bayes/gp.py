@@ -16,9 +16,12 @@ class GP(BaseAsyncBO):
     def __init__(self, num_warmup_trials=15, random_fraction=0.33,
                  acq_fun="EI", acq_n_points=10000, xi=0.01, kappa=1.96,
                  async_strategy="impute", imputed_metric="cl_min",
+                 interim_results=False, interim_results_interval=3,
                  pruner=None, pruner_kwargs=None):
         super().__init__(num_warmup_trials=num_warmup_trials,
                          random_fraction=random_fraction,
+                         interim_results=interim_results,
+                         interim_results_interval=interim_results_interval,
                          pruner=pruner, pruner_kwargs=pruner_kwargs)
         if acq_fun not in ("EI", "PI", "LCB"):
             raise ValueError("acq_fun must be EI, PI or LCB")
@@ -38,7 +41,8 @@ class GP(BaseAsyncBO):
             Matern,
         )
 
-        dim = len(self.searchspace.keys())
+        dim = len(self.searchspace.keys()) + (
+            1 if self.interim_results else 0)
         kernel = ConstantKernel(1.0) * Matern(
             length_scale=np.ones(dim), nu=2.5)
         return GaussianProcessRegressor(
@@ -70,6 +74,10 @@ class GP(BaseAsyncBO):
             return None
         dim = len(self.searchspace.keys())
         cand = np.random.uniform(0.0, 1.0, size=(self.acq_n_points, dim))
+        if self.interim_results:
+            # evaluate candidates at full progress
+            cand = np.concatenate(
+                [cand, np.ones((self.acq_n_points, 1))], axis=1)
         if self.async_strategy == "asy_ts":
             # asynchronous Thompson sampling: one posterior draw, minimize
             sample = model.sample_y(cand, n_samples=1,
@@ -81,7 +89,7 @@ class GP(BaseAsyncBO):
             score = self._acquisition(mu, sigma, y_best)
             best = cand[int(np.argmax(score))]
         values = self.searchspace.inverse_transform(
-            best.tolist(), normalize_categorical=True)
+            best[:dim].tolist(), normalize_categorical=True)
         return self.searchspace.list_to_dict(values)
 
     def _acquisition(self, mu, sigma, y_best):

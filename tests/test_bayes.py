@@ -94,3 +94,43 @@ def test_registry_strings():
     assert isinstance(resolve_controller("tpe", d), TPE)
     with pytest.raises(ValueError):
         resolve_controller("bogus", d)
+
+
+def test_gp_interim_results():
+    """GP with BOHB-style interim augmentation: heartbeat metrics at
+    fractional progress feed the surrogate alongside finals."""
+    np.random.seed(3)
+    import random
+
+    random.seed(3)
+    sp = Searchspace(x=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 30, direction="min")
+    opt = resolve_controller(
+        GP(num_warmup_trials=8, random_fraction=0.1,
+           interim_results=True, interim_results_interval=2), d)
+    opt._initialize()
+    finished = None
+    n = 0
+    while True:
+        t = opt.get_suggestion(finished)
+        if t is None:
+            break
+        n += 1
+        assert n <= 60
+        d._trial_store[t.trial_id] = t
+        # simulate a metric curve converging to the quadratic objective
+        final = (t.params["x"] - 0.4) ** 2
+        for s in range(6):
+            t.append_metric({"value": final + (6 - s) * 0.1, "step": s})
+        t.status = Trial.FINALIZED
+        t.final_metric = final
+        d._final_store.append(t)
+        del d._trial_store[t.trial_id]
+        finished = t
+    assert len(d._final_store) == 30
+    best = min(t.final_metric for t in d._final_store)
+    assert best < 0.05
+    # surrogate trained on augmented rows: D+1 columns
+    X, y, n_fin = opt.get_XY()
+    assert X.shape[1] == 2
+    assert X.shape[0] > n_fin  # interim rows present
