@@ -1,0 +1,66 @@
+"""Dynamic-batching model server tests (CPU)."""
+import threading
+
+import torch
+
+from maggy_amd.models import MLP
+from maggy_amd.serving import ModelServer, make_app
+
+
+def test_batching_and_results():
+    model = MLP(in_features=8, hidden=16, num_classes=3)
+    model.eval()
+    with ModelServer(model, max_batch=16, max_wait_ms=20) as server:
+        xs = [torch.randn(8) for _ in range(40)]
+        futs = [server.submit(x) for x in xs]
+        outs = [f.result(timeout=10) for f in futs]
+    with torch.no_grad():
+        ref = model(torch.stack(xs))
+    for o, r in zip(outs, ref):
+        torch.testing.assert_close(o, r, rtol=1e-5, atol=1e-6)
+    # batching actually happened (40 requests in < 40 batches)
+    assert server.stats["requests"] == 40
+    assert server.stats["batches"] < 40
+
+
+def test_concurrent_clients():
+    server = ModelServer(lambda b: b * 2.0, max_batch=8,
+                         max_wait_ms=5).start()
+    results = {}
+
+    def client(i):
+        results[i] = server.predict(torch.tensor([float(i)]))
+
+    threads = [threading.Thread(target=client, args=(i,))
+               for i in range(20)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    server.stop()
+    for i in range(20):
+        assert float(results[i]) == 2.0 * i
+
+
+def test_predict_fn_error_propagates():
+    def boom(batch):
+        raise RuntimeError("bad model")
+
+    with ModelServer(boom, max_wait_ms=1) as server:
+        fut = server.submit(torch.zeros(2))
+        import pytest
+
+        with pytest.raises(RuntimeError):
+            fut.result(timeout=10)
+
+
+def test_http_surface():
+    from fastapi.testclient import TestClient
+
+    with ModelServer(lambda b: b + 1.0, max_wait_ms=1) as server:
+        app = make_app(server)
+        client = TestClient(app)
+        r = client.post("/predict", json={"input": [1.0, 2.0]})
+        assert r.status_code == 200
+        assert r.json()["output"] == [2.0, 3.0]
+        assert client.get("/stats").json()["requests"] >= 1
