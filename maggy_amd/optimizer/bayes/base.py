@@ -118,6 +118,8 @@ class BaseAsyncBO(AbstractOptimizer):
             new_trial = self.create_trial(
                 params, sample_type="promoted", run_budget=budget,
                 parent_trial_id=parent_id)
+        if "bracket" in decision:
+            new_trial.info_dict["hb_bracket"] = decision["bracket"]
         self.pruner.report_trial(
             original_trial_id=parent_id, new_trial_id=new_trial.trial_id)
         return new_trial
@@ -137,10 +139,9 @@ class BaseAsyncBO(AbstractOptimizer):
     def on_resume(self, finalized):
         """Experiment resume: the surrogate trains on the preloaded
         final_store; shrink the warmup buffer accordingly and rebuild the
-        model."""
+        model.  With a pruner the bracket state is rebuilt too."""
         if self.pruner is not None:
-            raise NotImplementedError(
-                "resume with a pruner is not supported")
+            self.pruner.on_resume(finalized)
         del self.warmup_buffer[:len(finalized)]
         try:
             self.update_model(budget=0)

@@ -50,6 +50,8 @@ class RandomSearch(AbstractOptimizer):
             new_trial = self.create_trial(
                 params, sample_type="promoted", run_budget=budget,
                 parent_trial_id=parent_id)
+        if "bracket" in decision:
+            new_trial.info_dict["hb_bracket"] = decision["bracket"]
         self.pruner.report_trial(
             original_trial_id=parent_id, new_trial_id=new_trial.trial_id)
         return new_trial
@@ -64,10 +66,11 @@ class RandomSearch(AbstractOptimizer):
 
     def on_resume(self, finalized):
         """Experiment resume: skip as many pre-sampled configs as have
-        already finalized (random search has no state beyond the buffer)."""
+        already finalized; with a pruner, the pruner rebuilds its bracket
+        state from the persisted records."""
         if self.pruner is not None:
-            raise NotImplementedError(
-                "resume with a pruner is not supported")
+            self.pruner.on_resume(finalized)
+            return
         del self.config_buffer[:len(finalized)]
 
     def finalize_experiment(self, trials):

@@ -146,6 +146,7 @@ class Hyperband(AbstractPruner):
             run = br.next_run()
             if run is not None:
                 self._updating = br.bracket_id
+                run["bracket"] = br.bracket_id
                 return run
         if self.n_pending > 0:
             self._start_next_bracket()
@@ -165,3 +166,53 @@ class Hyperband(AbstractPruner):
 
     def num_trials(self):
         return sum(sum(br.n_configs) for br in self.brackets)
+
+    def on_resume(self, finalized):
+        """Rebuild bracket state from persisted trials (info_dict carries
+        hb_bracket; budgets map to rungs; parent_trial_id links promoted
+        slots).  Rungs are replayed bottom-up: a completed rung re-runs
+        the deterministic promotion to regenerate the next rung's slot
+        originals, then restored trials claim their slots."""
+        by_bracket = {}
+        for t in finalized:
+            b = (t.info_dict or {}).get("hb_bracket")
+            if b is None:
+                continue
+            by_bracket.setdefault(int(b), []).append(t)
+        for bid, trials in sorted(by_bracket.items()):
+            br = self.brackets[bid]
+            if br.state == _Bracket.INIT:
+                br.state = _Bracket.RUNNING
+                self.n_pending -= 1
+            by_rung = {}
+            for t in trials:
+                budget = int(t.params.get("budget", br.budgets[0]))
+                by_rung.setdefault(br.budgets.index(budget), []).append(t)
+            for rung in range(br.n_rungs):
+                restored = by_rung.get(rung, [])
+                if rung == 0:
+                    for t in restored:
+                        br.slots[0].append(
+                            {"original": t.trial_id, "actual": t.trial_id})
+                    br.started[0] = len(br.slots[0])
+                else:
+                    if not restored and not br._rung_complete():
+                        break
+                    if br._rung_complete() and br.rung == rung - 1:
+                        br._promote()  # regenerates this rung's originals
+                    parents = {(t.info_dict or {}).get("parent_trial_id"):
+                               t for t in restored}
+                    for slot in br.slots[rung]:
+                        t = parents.get(slot["original"])
+                        if t is not None:
+                            slot["actual"] = t.trial_id
+                    br.started[rung] = sum(
+                        1 for s in br.slots[rung] if s["actual"])
+                if not by_rung.get(rung):
+                    break
+            if br.rung == br.n_rungs - 1 and br._rung_complete():
+                br.state = _Bracket.FINISHED
+        # make sure at least one bracket is active when work remains
+        if not self.finished() and not any(
+                br.state == _Bracket.RUNNING for br in self.brackets):
+            self._start_next_bracket()
