@@ -59,11 +59,12 @@ def precompute_rope(dim, max_seq_len, theta):
     return torch.cos(freqs), torch.sin(freqs)
 
 
-def apply_rope(x, cos, sin):
-    # x: [B, H, T, D]; rotate pairs (x0,x1) in the last dim
+def apply_rope(x, cos, sin, pos=0):
+    # x: [B, H, T, D]; rotate pairs (x0,x1) in the last dim; ``pos`` is
+    # the absolute position of x's first token (KV-cached decode)
     T = x.shape[-2]
-    cos = cos[:T].to(x.dtype)
-    sin = sin[:T].to(x.dtype)
+    cos = cos[pos:pos + T].to(x.dtype)
+    sin = sin[pos:pos + T].to(x.dtype)
     x1, x2 = x[..., 0::2], x[..., 1::2]
     out = torch.empty_like(x)
     out[..., 0::2] = x1 * cos - x2 * sin
@@ -84,15 +85,25 @@ class Attention(nn.Module):
                             bias=False)
         self.wo = nn.Linear(cfg.n_heads * self.head_dim, cfg.dim, bias=False)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, cache=None, pos=0):
         B, T, _ = x.shape
         q = self.wq(x).view(B, T, self.n_heads, self.head_dim).transpose(1, 2)
         k = self.wk(x).view(B, T, self.n_kv_heads, self.head_dim).transpose(1, 2)
         v = self.wv(x).view(B, T, self.n_kv_heads, self.head_dim).transpose(1, 2)
-        q = apply_rope(q, cos, sin)
-        k = apply_rope(k, cos, sin)
+        q = apply_rope(q, cos, sin, pos)
+        k = apply_rope(k, cos, sin, pos)
+        if cache is not None:
+            # cache: dict with "k"/"v" [B, KV, T_past, D] (decode path)
+            if cache.get("k") is not None:
+                k = torch.cat([cache["k"], k], dim=2)
+                v = torch.cat([cache["v"], v], dim=2)
+            cache["k"], cache["v"] = k, v
+        # causal masking is needed only when the query block spans >1 new
+        # position; a single decoded token attends to the whole cache
+        causal = T > 1
         out = F.scaled_dot_product_attention(
-            q, k, v, is_causal=True, enable_gqa=(self.n_kv_heads != self.n_heads)
+            q, k, v, is_causal=causal,
+            enable_gqa=(self.n_kv_heads != self.n_heads)
         )
         out = out.transpose(1, 2).reshape(B, T, -1)
         return self.wo(out)
@@ -117,8 +128,8 @@ class Block(nn.Module):
         self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.ffn = FeedForward(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
+    def forward(self, x, cos, sin, cache=None, pos=0):
+        x = x + self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
         x = x + self.ffn(self.ffn_norm(x))
         return x
 
@@ -161,3 +172,36 @@ class LlamaModel(nn.Module):
 
     def num_params(self):
         return sum(p.numel() for p in self.parameters())
+
+    @torch.no_grad()
+    def generate(self, tokens, max_new_tokens, temperature=0.0, top_k=None):
+        """KV-cached autoregressive decoding (greedy at temperature 0).
+
+        :param tokens: [B, T_prompt] prompt ids
+        :returns: [B, T_prompt + max_new_tokens] ids
+        """
+        self.eval()
+        caches = [{"k": None, "v": None} for _ in self.layers]
+        out = tokens
+        x_in = tokens
+        pos = 0
+        for _ in range(max_new_tokens):
+            x = self.tok_emb(x_in)
+            for layer, cache in zip(self.layers, caches):
+                x = layer(x, self.rope_cos, self.rope_sin, cache=cache,
+                          pos=pos)
+            logits = self.lm_head(self.norm(x[:, -1:, :]))[:, -1, :]
+            if temperature > 0:
+                logits = logits / temperature
+                if top_k is not None:
+                    kth = torch.topk(logits, top_k, dim=-1).values[..., -1:]
+                    logits = logits.masked_fill(logits < kth,
+                                                float("-inf"))
+                probs = torch.softmax(logits.float(), dim=-1)
+                nxt = torch.multinomial(probs, 1)
+            else:
+                nxt = logits.argmax(dim=-1, keepdim=True)
+            pos += x_in.shape[1]
+            out = torch.cat([out, nxt], dim=1)
+            x_in = nxt
+        return out
