@@ -64,3 +64,21 @@ def test_http_surface():
         assert r.status_code == 200
         assert r.json()["output"] == [2.0, 3.0]
         assert client.get("/stats").json()["requests"] >= 1
+
+
+def test_generation_serving():
+    """Batched prompt-completion through the server (tiny Llama)."""
+    from maggy_amd.models import LlamaConfig, LlamaModel
+
+    torch.manual_seed(0)
+    model = LlamaModel(LlamaConfig.tiny(vocab_size=50)).eval()
+
+    def complete(batch):  # batch: [B, T] prompts -> [B, T+4] ids
+        return model.generate(batch, max_new_tokens=4)
+
+    with ModelServer(complete, max_batch=8, max_wait_ms=10) as server:
+        prompts = [torch.randint(0, 50, (6,)) for _ in range(12)]
+        outs = [server.submit(p).result(timeout=30) for p in prompts]
+    for p, o in zip(prompts, outs):
+        assert o.shape == (10,)
+        assert torch.equal(o[:6], p)
