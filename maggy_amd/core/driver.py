@@ -281,8 +281,18 @@ class OptimizationDriver:
                 trial.status = Trial.ERROR
             self._error_store.append(trial)
             self._trial_store.pop(trial_id, None)
+            self._notify_controller_error(trial)
         w.trial_id = None
         self._assign_next(w)
+
+    def _notify_controller_error(self, trial):
+        """Tell the controller a trial died so budget-based schedulers
+        (Hyperband brackets) free the slot instead of waiting forever."""
+        try:
+            if hasattr(self.controller, "on_trial_error"):
+                self.controller.on_trial_error(trial)
+        except Exception as e:
+            self.log("controller on_trial_error failed: {}".format(e))
 
     def _handle_metrics(self, records):
         """Metric-stream digestion + early-stop policy (parity
@@ -335,6 +345,7 @@ class OptimizationDriver:
                             trial.status = Trial.ERROR
                         self._error_store.append(trial)
                         self._trial_store.pop(lost_trial_id, None)
+                        self._notify_controller_error(trial)
                 w.trial_id = None
                 continue
             self.pool.respawn(w)

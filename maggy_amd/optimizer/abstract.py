@@ -86,6 +86,17 @@ class AbstractOptimizer(ABC):
     def finalize_experiment(self, trials):
         """Called once after the last trial finalized."""
 
+    def on_trial_error(self, trial):
+        """Driver callback when an assigned trial errors (user exception or
+        worker crash give-up).  Default: notify the pruner so its bracket
+        slot is reset/failed instead of stalling the rung (an errored trial
+        otherwise never reaches final_store and Hyperband's
+        ``_rung_complete`` stays false forever).  Optimizers without budget
+        state need no action — the driver simply asks for the next
+        suggestion."""
+        if self.pruner is not None and hasattr(self.pruner, "on_trial_error"):
+            self.pruner.on_trial_error(trial.trial_id)
+
     # -- helpers shared by concrete optimizers ---------------------------
     def create_trial(self, hparams, sample_type="random", run_budget=0,
                      model_budget=None, parent_trial_id=None):

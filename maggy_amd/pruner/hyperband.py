@@ -29,10 +29,14 @@ class _Bracket:
         self.n_rungs = len(n_configs)
         self.rung = 0
         self.started = [0] * self.n_rungs  # slots handed to the optimizer
-        # rung -> list of {"original": id, "actual": id|None}
+        # rung -> list of {"original": id, "actual": id|None[, "failed",
+        #                  "errors"]}
         self.slots = {r: [] for r in range(self.n_rungs)}
         self.metric_getter = metric_getter
         self._log = log
+        # bounded retry budget so a deterministically-failing train_fn
+        # cannot loop the bracket forever
+        self.error_retries_left = sum(n_configs)
 
     def next_run(self):
         """A schedulable run in this bracket, or None (busy/finished)."""
@@ -56,10 +60,13 @@ class _Bracket:
         return None
 
     def _rung_complete(self):
-        """Every slot of the current rung created AND finished."""
+        """Every slot of the current rung created AND finished (a slot
+        permanently marked failed counts as finished with worst-case)."""
         if len(self.slots[self.rung]) < self.n_configs[self.rung]:
             return False
         for slot in self.slots[self.rung]:
+            if slot.get("failed"):
+                continue
             if slot["actual"] is None:
                 return False
             if not self.metric_getter([slot["actual"]]):
@@ -67,15 +74,57 @@ class _Bracket:
         return True
 
     def _promote(self):
-        ids = [s["actual"] for s in self.slots[self.rung]]
-        metrics = self.metric_getter(ids)  # min convention
+        ids = [s["actual"] for s in self.slots[self.rung]
+               if not s.get("failed")]
+        metrics = {k: v for k, v in self.metric_getter(ids).items()
+                   if v is not None}
         ranked = sorted(metrics, key=metrics.get)
         keep = ranked[: self.n_configs[self.rung + 1]]
         self._log("bracket {} rung {} -> promote {}".format(
             self.bracket_id, self.rung, keep))
         self.rung += 1
+        # failed slots shrink the promotion pool; shrink the rung quota so
+        # _rung_complete can still be reached
+        if len(keep) < self.n_configs[self.rung]:
+            self.n_configs[self.rung] = len(keep)
+        if not keep and self.rung >= 1:
+            self.state = _Bracket.FINISHED
+            self._log("bracket {} finished (no promotable trials)".format(
+                self.bracket_id))
+            return
         for tid in keep:
             self.slots[self.rung].append({"original": tid, "actual": None})
+
+    def on_trial_error(self, trial_id):
+        """An assigned trial errored (user exception or worker give-up):
+        free its slot so the rung can complete instead of stalling the
+        bracket forever.  While the bracket's retry budget lasts, the slot
+        is reset to re-run (rung 0 re-samples a fresh config; higher rungs
+        re-run the promoted parent); once exhausted, the slot is marked
+        failed = finished-with-worst-case and excluded from promotion."""
+        for rung in range(self.n_rungs - 1, -1, -1):
+            for slot in self.slots[rung]:
+                if slot["actual"] == trial_id and not slot.get("failed"):
+                    if self.error_retries_left > 0:
+                        self.error_retries_left -= 1
+                        if rung == 0:
+                            self.slots[0].remove(slot)
+                        else:
+                            slot["actual"] = None
+                        self.started[rung] -= 1
+                        self._log(
+                            "bracket {} rung {}: trial {} errored; slot "
+                            "reset for re-run ({} retries left)".format(
+                                self.bracket_id, rung, trial_id,
+                                self.error_retries_left))
+                    else:
+                        slot["failed"] = True
+                        self._log(
+                            "bracket {} rung {}: trial {} errored; retry "
+                            "budget exhausted, slot marked failed".format(
+                                self.bracket_id, rung, trial_id))
+                    return True
+        return False
 
     def report(self, original_trial_id, new_trial_id):
         if self.rung == 0:
@@ -160,6 +209,17 @@ class Hyperband(AbstractPruner):
             self.brackets[self._updating].report(
                 original_trial_id, new_trial_id)
             self._updating = None
+
+    def on_trial_error(self, trial_id):
+        """Free the errored trial's bracket slot (see _Bracket.on_trial_error)
+        so pruning_routine can hand out a replacement instead of returning
+        IDLE forever."""
+        for br in self.brackets:
+            if br.state == _Bracket.FINISHED:
+                continue
+            if br.on_trial_error(trial_id):
+                return True
+        return False
 
     def finished(self):
         return all(br.state == _Bracket.FINISHED for br in self.brackets)

@@ -92,6 +92,17 @@ def continuation_fn(hparams, reporter, trial_dir, parent_checkpoint):
     return float(steps_done)
 
 
+def fails_once_fn(hparams, reporter):
+    """Budget-trial that raises on the first-ever invocation (sentinel file
+    under MAGGY_LOG_DIR survives across trials) — exercises the Hyperband
+    errored-slot re-run path."""
+    sentinel = os.path.join(os.environ["MAGGY_LOG_DIR"], "failed_once")
+    if not os.path.exists(sentinel):
+        open(sentinel, "w").close()
+        raise RuntimeError("deliberate first-trial failure")
+    return budgeted_fn(hparams, reporter)
+
+
 def noisy_quadratic_fn(hparams, reporter):
     """Deterministic objective for pool BO tests: peak at lr=0.06."""
     v = 1.0 - (hparams["lr"] - 0.06) ** 2 * 100.0

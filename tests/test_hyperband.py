@@ -112,3 +112,124 @@ def test_hyperband_idle_when_busy():
     t = opt.get_suggestion(started[-1])
     assert isinstance(t, Trial)
     assert t.params["budget"] == 2
+
+
+def _finish(d, t, metric):
+    t.status = Trial.FINALIZED
+    t.final_metric = metric
+    d._final_store.append(t)
+    d._trial_store.pop(t.trial_id, None)
+
+
+def test_hyperband_errored_trial_is_rerun():
+    """An errored trial frees its slot for a re-run instead of stalling the
+    bracket."""
+    random.seed(2)
+    np.random.seed(2)
+    sp = Searchspace(lr=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 100, direction="min")
+    opt = resolve_controller(
+        RandomSearch(pruner="hyperband",
+                     pruner_kwargs=dict(min_budget=1, max_budget=4, eta=2,
+                                        n_iterations=1)), d)
+    opt._initialize()
+    # bracket 0: n_configs [4, 2, 1], budgets [1, 2, 4]
+    first = opt.get_suggestion()
+    d._trial_store[first.trial_id] = first
+    # the first rung-0 trial errors
+    first.status = Trial.ERROR
+    d._trial_store.pop(first.trial_id, None)
+    opt.on_trial_error(first)
+    # the bracket must still complete: 7 successful trials total
+    finished = None
+    n = 0
+    budgets_run = []
+    while True:
+        t = opt.get_suggestion(finished)
+        if t is None:
+            break
+        assert t != "IDLE", "bracket stalled after an errored trial"
+        n += 1
+        assert n <= 20
+        budgets_run.append(t.params["budget"])
+        d._trial_store[t.trial_id] = t
+        _finish(d, t, t.params["lr"])
+        finished = t
+    assert n == 7
+    assert budgets_run.count(1) == 4
+    assert opt.pruner.finished()
+
+
+def test_hyperband_error_in_promoted_rung():
+    """An error in a promoted (rung>0) trial re-runs the same parent."""
+    random.seed(3)
+    np.random.seed(3)
+    sp = Searchspace(lr=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 100, direction="min")
+    opt = resolve_controller(
+        RandomSearch(pruner="hyperband",
+                     pruner_kwargs=dict(min_budget=1, max_budget=4, eta=2,
+                                        n_iterations=1)), d)
+    opt._initialize()
+    finished = None
+    for _ in range(4):  # complete rung 0
+        t = opt.get_suggestion(finished)
+        d._trial_store[t.trial_id] = t
+        _finish(d, t, t.params["lr"])
+        finished = t
+    promo = opt.get_suggestion(finished)
+    assert promo.params["budget"] == 2
+    parent = promo.info_dict["parent_trial_id"]
+    promo.status = Trial.ERROR
+    opt.on_trial_error(promo)
+    # same parent must be re-handed
+    retry = opt.get_suggestion()
+    assert retry.params["budget"] == 2
+    assert retry.info_dict["parent_trial_id"] == parent
+    d._trial_store[retry.trial_id] = retry
+    _finish(d, retry, retry.params["lr"])
+    finished = retry
+    n = 1
+    while True:
+        t = opt.get_suggestion(finished)
+        if t is None:
+            break
+        assert t != "IDLE"
+        n += 1
+        assert n <= 20
+        d._trial_store[t.trial_id] = t
+        _finish(d, t, t.params["lr"])
+        finished = t
+    assert opt.pruner.finished()
+
+
+def test_hyperband_retry_budget_exhausted_marks_failed():
+    """When the bracket's retry budget is spent, errored slots count as
+    finished-with-worst-case so rungs still complete (never hang)."""
+    random.seed(4)
+    np.random.seed(4)
+    sp = Searchspace(lr=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 100, direction="min")
+    opt = resolve_controller(
+        RandomSearch(pruner="hyperband",
+                     pruner_kwargs=dict(min_budget=1, max_budget=4, eta=2,
+                                        n_iterations=1)), d)
+    opt._initialize()
+    br = opt.pruner.brackets[0]
+    # every trial errors: retry budget = sum(n_configs) = 7, after which
+    # slots are marked failed
+    n_err = 0
+    while not opt.pruner.finished():
+        t = opt.get_suggestion()
+        if t is None:
+            break
+        assert t != "IDLE", "stalled with all-erroring train_fn"
+        d._trial_store[t.trial_id] = t
+        t.status = Trial.ERROR
+        d._trial_store.pop(t.trial_id, None)
+        opt.on_trial_error(t)
+        n_err += 1
+        assert n_err <= 40
+    # rung 0 completed via failed slots; nothing promotable -> FINISHED
+    assert br.state == "FINISHED"
+    assert opt.get_suggestion() is None

@@ -33,3 +33,19 @@ def test_hyperband_pool_e2e(exp_dir):
     assert sorted(budgets) == [1, 1, 1, 1, 2, 2, 4]
     # pruner.log artifact exists
     assert os.path.exists(os.path.join(run_dir, "pruner.log"))
+
+
+def test_hyperband_pool_survives_trial_error(exp_dir):
+    """A train_fn that raises must not stall its bracket: the slot is freed
+    and re-run, and the experiment completes."""
+    sp = Searchspace(lr=("DOUBLE", [0.01, 0.1]))
+    opt = RandomSearch(pruner="hyperband",
+                       pruner_kwargs=dict(min_budget=1, max_budget=4,
+                                          eta=2, n_iterations=1))
+    cfg = HyperparameterOptConfig(
+        num_trials=7, optimizer=opt, searchspace=sp, direction="max",
+        es_policy="none", num_workers=2, name="hb-err")
+    res = experiment.lagom(fns.fails_once_fn, cfg)
+    # 7 successful trials despite the first one erroring
+    assert res["num_trials"] == 7
+    assert opt.pruner.finished()
