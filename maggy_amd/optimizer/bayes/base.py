@@ -196,9 +196,19 @@ class BaseAsyncBO(AbstractOptimizer):
                     X.append(xb + [(s + 1) / T])
                     y.append(sign * hist[s])
         if include_busy and n_fin:
+            # kriging believer imputes each busy location with the current
+            # model's posterior mean (reference gp.py:329-373); constant
+            # liar uses a fixed pessimistic/optimistic value.  KB falls
+            # back to cl_min before the first model fit.
+            kb_model = None
+            if self.imputed_metric == "kb":
+                kb_model = self.models.get(budget)
+                if kb_model is not None and not hasattr(
+                        kb_model, "X_train_"):
+                    kb_model = None
             liar = {
                 "cl_min": min(y[:n_fin]), "cl_max": max(y[:n_fin]),
-                "cl_mean": sum(y[:n_fin]) / n_fin,
+                "cl_mean": sum(y[:n_fin]) / n_fin, "kb": min(y[:n_fin]),
             }[self.imputed_metric]
             for t in self.trial_store.values():
                 params = {k: v for k, v in t.params.items()
@@ -212,5 +222,9 @@ class BaseAsyncBO(AbstractOptimizer):
                 if self.interim_results:
                     x = x + [1.0]
                 X.append(x)
-                y.append(liar)
+                if kb_model is not None:
+                    y.append(float(kb_model.predict(
+                        np.asarray([x], dtype=float))[0]))
+                else:
+                    y.append(liar)
         return np.asarray(X, dtype=float), np.asarray(y, dtype=float), n_fin

@@ -134,3 +134,100 @@ def test_gp_interim_results():
     X, y, n_fin = opt.get_XY()
     assert X.shape[1] == 2
     assert X.shape[0] > n_fin  # interim rows present
+
+
+def test_gp_incumbent_original_scale():
+    """EI's incumbent must be the ORIGINAL-scale y minimum, not sklearn's
+    normalized y_train_: with a shifted objective
+    (metric ~ +1000) the acquisition must still optimize."""
+    np.random.seed(4)
+    import random
+
+    random.seed(4)
+    sp = Searchspace(x=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 30, direction="min")
+    opt = resolve_controller(GP(num_warmup_trials=8, random_fraction=0.1), d)
+    opt._initialize()
+    finals = run_sequential(
+        opt, d, lambda p: 1000.0 + (p["x"] - 0.6) ** 2)
+    # incumbent bookkeeping is original scale
+    assert abs(opt._y_fit_min[0] - min(t.final_metric
+                                       for t in finals)) < 1e-9
+    assert min(t.final_metric for t in finals) < 1000.005
+    assert any(t.info_dict.get("sample_type") == "model" for t in finals)
+
+
+def test_gp_kriging_believer():
+    """Kriging-believer imputation: busy locations get the model's own
+    posterior mean, not a constant (reference gp.py:329-373)."""
+    np.random.seed(5)
+    import random
+
+    random.seed(5)
+    sp = Searchspace(x=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 20, direction="min")
+    opt = resolve_controller(
+        GP(num_warmup_trials=6, random_fraction=0.1,
+           imputed_metric="kb"), d)
+    opt._initialize()
+    # finish 6 warmup trials so a model exists
+    finished = None
+    for _ in range(6):
+        t = opt.get_suggestion(finished)
+        d._trial_store[t.trial_id] = t
+        t.status = Trial.FINALIZED
+        t.final_metric = (t.params["x"] - 0.5) ** 2
+        d._final_store.append(t)
+        del d._trial_store[t.trial_id]
+        finished = t
+    opt.update_model(0)
+    # park a busy trial at x=0.5 (the optimum): KB should impute a value
+    # near 0, far from cl_min-of-warmup only if warmup hit near 0.5
+    busy = Trial({"x": 0.5})
+    d._trial_store[busy.trial_id] = busy
+    X, y, n_fin = opt.get_XY(include_busy=True)
+    assert X.shape[0] == n_fin + 1
+    kb_value = y[-1]
+    model_mu = float(opt.models[0].predict(np.asarray([[0.5]]))[0])
+    assert abs(kb_value - model_mu) < 1e-9
+
+
+def test_gp_invalid_imputed_metric():
+    with pytest.raises(ValueError):
+        GP(imputed_metric="bogus")
+
+
+def test_gp_lbfgs_refinement_improves_acquisition():
+    """The L-BFGS-B polish never returns a candidate with worse acquisition
+    than the best sampled point."""
+    np.random.seed(6)
+    sp = Searchspace(x=("DOUBLE", [0.0, 1.0]), y=("DOUBLE", [0.0, 1.0]))
+    d = FakeDriver(sp, 10, direction="min")
+    opt = resolve_controller(
+        GP(num_warmup_trials=6, random_fraction=0.0, acq_n_points=50), d)
+    opt._initialize()
+    finished = None
+    for _ in range(6):
+        t = opt.get_suggestion(finished)
+        d._trial_store[t.trial_id] = t
+        t.status = Trial.FINALIZED
+        t.final_metric = (t.params["x"] - 0.3) ** 2 + (t.params["y"]) ** 2
+        d._final_store.append(t)
+        del d._trial_store[t.trial_id]
+        finished = t
+    opt.update_model(0)
+    model = opt.models[0]
+    y_best = opt._y_fit_min[0]
+    np.random.seed(7)
+    cand = np.random.uniform(0, 1, size=(50, 2))
+    mu, sigma = model.predict(cand, return_std=True)
+    score = opt._acquisition(mu, sigma, y_best)
+    order = np.argsort(score)[::-1]
+    sampled_best = cand[int(order[0])]
+    refined = opt._refine_lbfgs(model, y_best, cand[order[:5]], sampled_best)
+
+    def acq(x):
+        m, s = model.predict(x.reshape(1, -1), return_std=True)
+        return float(opt._acquisition(m, s, y_best)[0])
+
+    assert acq(refined) >= acq(sampled_best) - 1e-12
