@@ -62,12 +62,35 @@ class MaggyParquetDataLoader:
                 rank, world_size = 0, 1
         self.rank = rank
         self.world_size = world_size
-        self.row_groups = list(
-            range(rank, self.file.num_row_groups, world_size))
+        self.row_groups = self._rank_groups(rank)
+        # Lockstep DDP requires EQUAL per-rank iteration counts: with
+        # num_row_groups % world_size != 0 (or uneven group sizes) a naive
+        # round-robin shard gives some ranks fewer batches and the others
+        # hang in all-reduce.  Every rank computes every
+        # rank's batch count from the (shared) file metadata and pads its
+        # own iteration by wrapping — the DistributedSampler convention —
+        # up to the global maximum.  No communication needed.
+        counts = [self._batch_count(self._rank_groups(r))
+                  for r in range(world_size)]
+        self.num_batches = max(counts) if counts else 0
         self._device = (torch.device("cuda", torch.cuda.current_device())
                         if torch.cuda.is_available() else None)
 
-    def __iter__(self):
+    def _rank_groups(self, rank):
+        n = self.file.num_row_groups
+        groups = list(range(rank, n, self.world_size))
+        if not groups and n:
+            groups = [rank % n]  # more ranks than row groups: wrap
+        return groups
+
+    def _batch_count(self, groups):
+        md = self.file.metadata
+        return sum(
+            (md.row_group(rg).num_rows + self.batch_size - 1)
+            // self.batch_size
+            for rg in groups)
+
+    def _iter_once(self):
         import numpy as np
 
         for rg in self.row_groups:
@@ -88,10 +111,19 @@ class MaggyParquetDataLoader:
                     batch[c] = t
                 yield batch
 
+    def __iter__(self):
+        yielded = 0
+        while yielded < self.num_batches:
+            for batch in self._iter_once():
+                yield batch
+                yielded += 1
+                if yielded >= self.num_batches:
+                    return
+            if yielded == 0:
+                return  # empty file: don't spin
+
     def __len__(self):
-        total = sum(self.file.metadata.row_group(rg).num_rows
-                    for rg in self.row_groups)
-        return (total + self.batch_size - 1) // self.batch_size
+        return self.num_batches
 
 
 class _DeviceIter:
@@ -126,10 +158,26 @@ class _DeviceIter:
     def __iter__(self):
         return self
 
+    def _record(self, batch, stream):
+        # the batch tensors were allocated on the copy stream; tell the
+        # caching allocator they are consumed on the compute stream, or
+        # their memory could be re-handed to the next _preload H2D copy
+        # while compute-stream kernels still read it
+        if torch.is_tensor(batch):
+            batch.record_stream(stream)
+        elif isinstance(batch, (list, tuple)):
+            for b in batch:
+                self._record(b, stream)
+        elif isinstance(batch, dict):
+            for b in batch.values():
+                self._record(b, stream)
+
     def __next__(self):
         if self._next is None:
             raise StopIteration
-        torch.cuda.current_stream().wait_stream(self.stream)
+        cur = torch.cuda.current_stream()
+        cur.wait_stream(self.stream)
         batch = self._next
+        self._record(batch, cur)
         self._preload()
         return batch
