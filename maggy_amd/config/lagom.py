@@ -1,8 +1,11 @@
 """Abstract experiment config.
 
 Parity: /root/reference/maggy/config/lagom.py:22-34 (name / description /
-hb_interval). ``hb_interval`` survives as the driver's metric-drain cadence
-upper bound; with the shared-memory reporter there is no heartbeat socket.
+hb_interval). ``hb_interval`` is VESTIGIAL: it is accepted for source
+compatibility with reference configs but has no effect — the reference used
+it as the executor heartbeat-socket period (rpc.py:716-737); here metrics
+stream through shared-memory rings drained every event-loop tick, so there
+is no heartbeat cadence to configure.
 """
 from abc import ABC
 

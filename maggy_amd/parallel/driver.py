@@ -46,6 +46,7 @@ class TorchDistributedTrainingDriver:
             "test_set": self.config.test_set,
             "hparams": self.config.hparams,
             "zero_lvl": self.config.zero_lvl,
+            "mixed_precision": self.config.mixed_precision,
             "bucket_cap_mb": self.config.bucket_cap_mb,
             "master_addr": "127.0.0.1",
             "master_port": random.randint(20000, 49000),

@@ -22,16 +22,31 @@ from maggy_amd.core.shm import MetricRing
 
 class DDPModuleWrapper:
     """Class factory: wraps a user module class so that instantiating it
-    inside train_fn yields a device-placed, DDP-wrapped module (parity:
-    MaggyDDPModuleWrapper, patching/modules.py:38-65)."""
+    inside train_fn yields a device-placed, distribution-wrapped module
+    (parity: MaggyDDPModuleWrapper, patching/modules.py:38-65).
+
+    zero_lvl 0/1: DDP (bucketed all-reduce over xGMI).  zero_lvl 2: no DDP
+    — gradients are reduce-scattered to their shard owner inside the
+    patched ZeRO optimizer's step(), so the module is only device-placed
+    and its initial parameters broadcast from rank 0."""
 
     @classmethod
-    def build(cls, module_cls, bucket_cap_mb=None):
+    def build(cls, module_cls, bucket_cap_mb=None, zero_lvl=0):
         class _Wrapped(module_cls):
             def __new__(wcls, *args, **kwargs):
                 inner = module_cls(*args, **kwargs)
                 if torch.cuda.is_available():
                     inner = inner.cuda()
+                if zero_lvl >= 2:
+                    import torch.distributed as dist
+
+                    from maggy_amd.parallel.zero import (
+                        broadcast_module_params,
+                    )
+
+                    if dist.is_initialized() and dist.get_world_size() > 1:
+                        broadcast_module_params(inner)
+                    return inner
                 from maggy_amd.parallel.dist import wrap_ddp
 
                 return wrap_ddp(inner, bucket_cap_mb=bucket_cap_mb)
@@ -54,12 +69,25 @@ def _patch_torch(zero_lvl):
 
         from maggy_amd.parallel.zero import ZeroFusedAdam, ZeroFusedSGD
 
+        if zero_lvl >= 2:
+            # level 2: gradient reduction is sharded too (no DDP all-reduce)
+            class _Adam(ZeroFusedAdam):
+                def __init__(self, params, **kw):
+                    kw.setdefault("grad_shard", True)
+                    super().__init__(params, **kw)
+
+            class _SGD(ZeroFusedSGD):
+                def __init__(self, params, **kw):
+                    kw.setdefault("grad_shard", True)
+                    super().__init__(params, **kw)
+        else:
+            _Adam, _SGD = ZeroFusedAdam, ZeroFusedSGD
         originals["Adam"] = topt.Adam
         originals["AdamW"] = topt.AdamW
         originals["SGD"] = topt.SGD
-        topt.Adam = ZeroFusedAdam
-        topt.AdamW = ZeroFusedAdam
-        topt.SGD = ZeroFusedSGD
+        topt.Adam = _Adam
+        topt.AdamW = _Adam
+        topt.SGD = _SGD
     return originals
 
 
@@ -114,11 +142,13 @@ def dist_worker_main(rank, world_size, gpu_id, conn, ring_name, ring_slots,
             torch.cuda.set_device(0)
 
         module_cls = payload["module"]
+        zero_lvl = payload.get("zero_lvl", 0)
         wrapped = None
         if module_cls is not None:
             wrapped = DDPModuleWrapper.build(
-                module_cls, bucket_cap_mb=payload.get("bucket_cap_mb"))
-        originals = _patch_torch(payload.get("zero_lvl", 0))
+                module_cls, bucket_cap_mb=payload.get("bucket_cap_mb"),
+                zero_lvl=zero_lvl)
+        originals = _patch_torch(zero_lvl)
 
         trial_dir = os.path.join(log_dir, "dist_run")
         os.makedirs(trial_dir, exist_ok=True)
@@ -143,7 +173,16 @@ def dist_worker_main(rank, world_size, gpu_id, conn, ring_name, ring_slots,
                        "rank": rank,
                        "world_size": world_size},
             )
-            retval = train_fn(**kwargs)
+            if payload.get("mixed_precision"):
+                # the reference's mixed_precision flag mapped to fairscale
+                # FSDP fp16 (modules.py:75-97); MI355X-native: bf16
+                # autocast around the whole training function
+                device_type = "cuda" if torch.cuda.is_available() else "cpu"
+                with torch.autocast(device_type=device_type,
+                                    dtype=torch.bfloat16):
+                    retval = train_fn(**kwargs)
+            else:
+                retval = train_fn(**kwargs)
             if rank == 0 and retval is not None:
                 util.handle_return_val(
                     retval, trial_dir,

@@ -102,3 +102,67 @@ def dist_crashing_fn(module, hparams, reporter):
         raise RuntimeError("deliberate rank failure")
     reporter.broadcast(1.0, 0)
     return 1.0
+
+
+def dist_zero2_fn(module, hparams, reporter):
+    """ZeRO-2 path: no DDP wrap (module returns the bare nn.Module), the
+    patched optimizer reduce-scatters gradients to shard owners, frees
+    non-owned grads, and broadcasts updated params."""
+    import torch.distributed as dist
+
+    torch.manual_seed(11 + dist.get_rank())
+    model = module(hidden=32)
+    # ZeRO-2 must NOT DDP-wrap
+    assert isinstance(model, torch.nn.Module)
+    assert not isinstance(model, torch.nn.parallel.DistributedDataParallel)
+    # initial params were broadcast from rank 0: verify sync before training
+    for p in model.parameters():
+        gathered = [torch.zeros_like(p) for _ in range(dist.get_world_size())]
+        dist.all_gather(gathered, p.detach())
+        for g in gathered:
+            assert torch.allclose(g, p.detach()), "initial params differ"
+    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    from maggy_amd.parallel.zero import ZeroFusedAdam
+
+    assert isinstance(opt, ZeroFusedAdam) and opt.grad_shard, type(opt)
+    torch.manual_seed(100 + dist.get_rank())  # per-rank batches differ
+    x = torch.randn(16, 8)
+    y = torch.randint(0, 2, (16,))
+    first = last = None
+    for step in range(8):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+        last = float(loss)
+        if first is None:
+            first = last
+        reporter.broadcast(last, step)
+    # after step(): non-owned grads were freed
+    freed = sum(1 for p in model.parameters() if p.grad is None)
+    assert freed > 0, "ZeRO-2 should free non-owned grads"
+    # replicas stayed in sync through reduce+broadcast
+    for p in model.parameters():
+        gathered = [torch.zeros_like(p) for _ in range(dist.get_world_size())]
+        dist.all_gather(gathered, p.detach())
+        for g in gathered:
+            assert torch.allclose(g, p.detach(), atol=1e-6), \
+                "zero-2 replicas diverged"
+    assert last < first, "training did not progress"
+    return last
+
+
+def dist_autocast_fn(module, hparams, reporter):
+    """mixed_precision=True must run the train function under bf16
+    autocast (round-1 ADVICE: the flag was a silent no-op)."""
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    assert torch.is_autocast_enabled(device_type), \
+        "mixed_precision did not enable autocast"
+    a = torch.randn(4, 4)
+    b = torch.randn(4, 4)
+    if device_type == "cuda":
+        a, b = a.cuda(), b.cuda()
+    out = a @ b
+    assert out.dtype == torch.bfloat16, out.dtype
+    reporter.broadcast(1.0, 0)
+    return 1.0

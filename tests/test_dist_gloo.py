@@ -41,3 +41,25 @@ def test_dist_rank_failure_raises(exp_dir):
         module=fns.TinyNet, num_gpus=2, name="crash-gloo")
     with pytest.raises(WorkerCrashError):
         experiment.lagom(fns.dist_crashing_fn, cfg)
+
+
+@pytest.mark.timeout(180)
+def test_zero2_grad_sharding_gloo(exp_dir):
+    cfg = TorchDistributedConfig(
+        module=fns.TinyNet, num_gpus=2, zero_lvl=2, name="zero2-gloo")
+    res = experiment.lagom(fns.dist_zero2_fn, cfg)
+    assert res["final_metric_avg"] is not None
+
+
+@pytest.mark.timeout(180)
+def test_mixed_precision_autocast_gloo(exp_dir):
+    cfg = TorchDistributedConfig(
+        module=fns.TinyNet, num_gpus=2, mixed_precision=True,
+        name="amp-gloo")
+    res = experiment.lagom(fns.dist_autocast_fn, cfg)
+    assert res["final_metric_avg"] == 1.0
+
+
+def test_zero_lvl_3_rejected():
+    with pytest.raises(ValueError, match="zero_lvl"):
+        TorchDistributedConfig(module=fns.TinyNet, zero_lvl=3)
