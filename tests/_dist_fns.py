@@ -166,3 +166,49 @@ def dist_autocast_fn(module, hparams, reporter):
     assert out.dtype == torch.bfloat16, out.dtype
     reporter.broadcast(1.0, 0)
     return 1.0
+
+
+class TinyLlama(torch.nn.Module):
+    """Llama-tiny wrapper with a kwargs ctor for the module-class contract."""
+
+    def __init__(self, vocab_size=256):
+        super().__init__()
+        from maggy_amd.models.llama import LlamaConfig, LlamaModel
+
+        self.inner = LlamaModel(LlamaConfig.tiny(vocab_size=vocab_size))
+
+    def forward(self, tokens, targets=None):
+        return self.inner(tokens, targets=targets)
+
+
+def dist_llama_fn(module, hparams, reporter):
+    """Llama-tiny DDP e2e: per-rank batches, replicas must stay in sync
+    (round-1 VERDICT #3: cover the DP path the driver will scale to 8
+    ranks)."""
+    import torch.distributed as dist
+
+    torch.manual_seed(3)  # identical init on all ranks
+    model = module(vocab_size=int(hparams.get("vocab", 256)))
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    rank = dist.get_rank()
+    torch.manual_seed(1000 + rank)  # different data per rank
+    x = torch.randint(0, 256, (2, 16))
+    dev = next(model.parameters()).device
+    x = x.to(dev)
+    first = last = None
+    for step in range(4):
+        opt.zero_grad()
+        loss = model(x, targets=x)
+        loss.backward()
+        opt.step()
+        last = float(loss)
+        if first is None:
+            first = last
+        reporter.broadcast(last, step)
+    p0 = next(model.parameters()).detach().clone()
+    gathered = [torch.zeros_like(p0) for _ in range(dist.get_world_size())]
+    dist.all_gather(gathered, p0)
+    for g in gathered:
+        assert torch.allclose(g, p0, atol=1e-6), "llama replicas diverged"
+    assert last < first
+    return last

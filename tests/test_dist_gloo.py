@@ -63,3 +63,16 @@ def test_mixed_precision_autocast_gloo(exp_dir):
 def test_zero_lvl_3_rejected():
     with pytest.raises(ValueError, match="zero_lvl"):
         TorchDistributedConfig(module=fns.TinyNet, zero_lvl=3)
+
+
+@pytest.mark.timeout(300)
+def test_llama_tiny_ddp_8rank_gloo(exp_dir):
+    """8-rank data-parallel Llama-tiny on CPU/gloo: the same world size the
+    driver's SCALE run uses on 8 GPUs."""
+    cfg = TorchDistributedConfig(
+        module=fns.TinyLlama, hparams={"vocab": 256}, num_gpus=8,
+        name="llama8-gloo")
+    res = experiment.lagom(fns.dist_llama_fn, cfg)
+    assert res["world_size"] == 8
+    assert len(res["per_rank"]) == 8
+    assert res["final_metric_avg"] is not None
