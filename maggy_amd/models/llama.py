@@ -15,7 +15,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from maggy_amd.ops.fused_rms import MaggyRMSNorm, swiglu
+from maggy_amd.ops.fused_rms import MaggyRMSNorm, swiglu  # noqa: F401
+from maggy_amd.ops.linear import MaggyFeedForward, MaggyLinear
 
 
 @dataclass
@@ -78,12 +79,13 @@ class Attention(nn.Module):
         self.n_heads = cfg.n_heads
         self.n_kv_heads = cfg.n_kv_heads
         self.head_dim = cfg.dim // cfg.n_heads
-        self.wq = nn.Linear(cfg.dim, cfg.n_heads * self.head_dim, bias=False)
-        self.wk = nn.Linear(cfg.dim, cfg.n_kv_heads * self.head_dim,
-                            bias=False)
-        self.wv = nn.Linear(cfg.dim, cfg.n_kv_heads * self.head_dim,
-                            bias=False)
-        self.wo = nn.Linear(cfg.n_heads * self.head_dim, cfg.dim, bias=False)
+        # MaggyLinear = nn.Linear running on the in-tree MFMA GEMM
+        # (fwd + both backward layouts) when the shapes tile; eager/rocBLAS
+        # fallback otherwise (tiny CPU configs)
+        self.wq = MaggyLinear(cfg.dim, cfg.n_heads * self.head_dim)
+        self.wk = MaggyLinear(cfg.dim, cfg.n_kv_heads * self.head_dim)
+        self.wv = MaggyLinear(cfg.dim, cfg.n_kv_heads * self.head_dim)
+        self.wo = MaggyLinear(cfg.n_heads * self.head_dim, cfg.dim)
 
     def forward(self, x, cos, sin, cache=None, pos=0):
         B, T, _ = x.shape
@@ -109,15 +111,12 @@ class Attention(nn.Module):
         return self.wo(out)
 
 
-class FeedForward(nn.Module):
-    def __init__(self, cfg):
-        super().__init__()
-        self.w1 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False)  # gate
-        self.w3 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False)  # up
-        self.w2 = nn.Linear(cfg.ffn_hidden, cfg.dim, bias=False)  # down
+class FeedForward(MaggyFeedForward):
+    """Llama MLP: custom-GEMM w1/w3 with the SwiGLU fused into the w3
+    GEMM's epilogue (ops/linear.py); eager fallback off-GPU."""
 
-    def forward(self, x):
-        return self.w2(swiglu(self.w1(x), self.w3(x)))
+    def __init__(self, cfg):
+        super().__init__(cfg.dim, cfg.ffn_hidden)
 
 
 class Block(nn.Module):
@@ -141,7 +140,7 @@ class LlamaModel(nn.Module):
         self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.dim)
         self.layers = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layers))
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
-        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self.lm_head = MaggyLinear(cfg.dim, cfg.vocab_size)
         cos, sin = precompute_rope(cfg.dim // cfg.n_heads, cfg.max_seq_len,
                                    cfg.rope_theta)
         self.register_buffer("rope_cos", cos, persistent=False)

@@ -20,6 +20,7 @@ SOURCES = [
     os.path.join(HIP_DIR, "maggy_kernels.hip"),
     os.path.join(HIP_DIR, "fused_bn.hip"),
     os.path.join(HIP_DIR, "fused_rms.hip"),
+    os.path.join(HIP_DIR, "gemm.hip"),
 ]
 
 

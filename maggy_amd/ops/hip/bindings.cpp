@@ -40,6 +40,13 @@ extern "C" void launch_reduce_sum_bf16(const void*, long long, float*,
                                        hipStream_t);
 extern "C" void launch_reduce_max_f32(const float*, long long, float*,
                                       hipStream_t);
+extern "C" void launch_gemm_tn_bf16(const void*, const void*, void*, int,
+                                    int, int, hipStream_t);
+extern "C" void launch_gemm_tn_bf16_swiglu(const void*, const void*, void*,
+                                           const void*, void*, int, int,
+                                           int, hipStream_t);
+extern "C" void launch_transpose_bf16(const void*, void*, int, int,
+                                      hipStream_t);
 
 namespace {
 
@@ -204,6 +211,60 @@ void reduce_max(torch::Tensor in, torch::Tensor out) {
                         out.data_ptr<float>(), current_stream());
 }
 
+void check_gemm_operand(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a CUDA tensor");
+  TORCH_CHECK(t.dtype() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.dim() == 2, name, " must be 2-D");
+}
+
+// C[M][N] = A[M][K] @ W[N][K]^T (torch F.linear layout), bf16, fp32 acc
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor w) {
+  check_gemm_operand(a, "a");
+  check_gemm_operand(w, "w");
+  const int64_t M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "gemm_tn: inner dims mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "gemm_tn needs M,N % 256 == 0 and K % 64 == 0; got ", M, "x",
+              N, "x", K);
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemm_tn_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(), (int)M,
+                      (int)N, (int)K, current_stream());
+  return c;
+}
+
+// y3 = A @ W3^T with fused h = silu(y1) * y3 epilogue; returns (y3, h)
+std::vector<torch::Tensor> gemm_tn_swiglu(torch::Tensor a, torch::Tensor w3,
+                                          torch::Tensor y1) {
+  check_gemm_operand(a, "a");
+  check_gemm_operand(w3, "w3");
+  check_gemm_operand(y1, "y1");
+  const int64_t M = a.size(0), K = a.size(1), N = w3.size(0);
+  TORCH_CHECK(w3.size(1) == K, "gemm_tn_swiglu: inner dims mismatch");
+  TORCH_CHECK(y1.size(0) == M && y1.size(1) == N,
+              "gemm_tn_swiglu: y1 must be [M][N]");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "gemm_tn_swiglu needs M,N % 256 == 0 and K % 64 == 0");
+  auto y3 = torch::empty({M, N}, a.options());
+  auto h = torch::empty({M, N}, a.options());
+  launch_gemm_tn_bf16_swiglu(a.data_ptr(), w3.data_ptr(), y3.data_ptr(),
+                             y1.data_ptr(), h.data_ptr(), (int)M, (int)N,
+                             (int)K, current_stream());
+  return {y3, h};
+}
+
+// out[c][r] = in[r][c], bf16, R and C multiples of 64
+torch::Tensor transpose2d(torch::Tensor in) {
+  check_gemm_operand(in, "in");
+  const int64_t R = in.size(0), C = in.size(1);
+  TORCH_CHECK(R % 64 == 0 && C % 64 == 0,
+              "transpose2d needs dims % 64 == 0; got ", R, "x", C);
+  auto out = torch::empty({C, R}, in.options());
+  launch_transpose_bf16(in.data_ptr(), out.data_ptr(), (int)R, (int)C,
+                        current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -223,4 +284,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd, "fused silu(g)*u backward");
   m.def("reduce_sum", &reduce_sum, "scalar sum reduction");
   m.def("reduce_max", &reduce_max, "scalar max reduction");
+  m.def("gemm_tn", &gemm_tn,
+        "bf16 MFMA GEMM C = A @ W^T (torch Linear layout)");
+  m.def("gemm_tn_swiglu", &gemm_tn_swiglu,
+        "bf16 MFMA GEMM y3 = A @ W3^T with fused h = silu(y1)*y3");
+  m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
 }
