@@ -9,11 +9,15 @@ config (q, kv, o, gate/up, down, lm_head).  For each: fwd (TN), dX
 cost counted against the custom path.
 """
 import argparse
+import os
+import sys
 import time
 
 import torch
 
-from maggy_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from maggy_amd import ops  # noqa: E402
 
 
 def bench(fn, iters=20, warmup=5):
