@@ -460,6 +460,209 @@ void gemm_tn_bf16_v3(const uint16_t* __restrict__ A,
   }
 }
 
+// ------------------------------------------------- v4: consumption-ordered
+// 8-phase pipeline (fixes the round-1 choreography hole in DESIGN_8phase.md).
+//
+// Keeps v1's 2Mx4N wave split.  The insight: per phase q of a tile, the
+// block consumes A rows [32q,32q+32) u [128+32q,+32) and (at q=0) the
+// whole W tile into registers — so define the A staging units as those
+// ROW SETS (u01 = rows [0,64)u[128,192), u23 = [64,128)u[192,256)), and
+// every 16 KiB staged unit becomes dead-before-stage by construction:
+//
+//   iteration i computes tiles a=2i (buf0), b=2i+1 (buf1); per phase:
+//     P0: stage buf1.A23 for tile b     (region last read P7 of i-1)
+//     P1: buf0.W.h0 (a+2)  P2: buf0.W.h1  P3: buf0.A01  P4: buf0.A23
+//     P5: buf1.W.h0 (b+2)  P6: buf1.W.h1  P7: buf1.A01
+//   waits: end of P3 and P7 only, vmcnt(6) (= 3 units in flight, 2 glds
+//   each); FIFO check: the P3 wait drains {P5,P6,P7 of i-1, P0 of i} =
+//   exactly tile b's four units; the P7 wait drains {P1..P4 of i} = tile
+//   (a+2)'s units before P0 of i+1 reads them.  Tail: when a+2/b+2 >= KT
+//   nothing is staged and the wait becomes vmcnt(0).  KT must be even.
+extern "C" __global__ __launch_bounds__(THREADS, 2)
+void gemm_tn_bf16_v4(const uint16_t* __restrict__ A,
+                     const uint16_t* __restrict__ W,
+                     uint16_t* __restrict__ C, int M, int N, int K) {
+  __shared__ uint16_t lds[2 * 2 * BM * BK];
+  const int nwgM = M / BM, nwgN = N / BN;
+  int wg = xcd_remap(blockIdx.x, nwgM * nwgN);
+  const int bm = (wg / nwgN) * BM;
+  const int bn = (wg % nwgN) * BN;
+  const int l = threadIdx.x;
+  const int wave = l >> 6;
+  const int lane = l & 63;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  f32x4 acc[MF][NF];
+  #pragma unroll
+  for (int i = 0; i < MF; ++i)
+    #pragma unroll
+    for (int j = 0; j < NF; ++j)
+      acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int KT = K / BK;
+
+  // Pre-swizzle staging offsets ONCE (hoisted out of the K-loop so the
+  // in-loop address math is one add — without this, register reuse of
+  // the address VGPRs makes the compiler insert vmcnt(0) WAR waits
+  // before the half-1 glds, defeating the counted-flight pipeline):
+  //   W pieces p = h*2+g : s = h*16384 + wave*2048 + g*1024
+  //   A pieces p = u*2+g : s = u*8192  + wave*1024 + g*16384
+  // For each: LDS element offset = s>>1 (wave-uniform base; hardware
+  // appends lane*16) and global ELEMENT offset within the tile
+  // (row*K + col) from the st_16x32 swizzle of s + lane*16.
+  // Only FOUR per-lane offsets are held across the loop: because the
+  // st_16x32 swizzle leaves bit 14 alone, swz(s + 16384) = swz(s) +
+  // 16384, i.e. a +16384-byte piece step is a +128-row (= +128*K
+  // element) global step — so every piece offset derives from {W g=0,
+  // W g=1, A u=0, A u=1} plus the wave-uniform 128*K.  The LDS
+  // destinations are wave-uniform (readfirstlane'd so they cost SGPRs,
+  // not VGPRs).  This keeps the kernel at <=256 VGPRs — with 16 hoisted
+  // offsets the accumulators spilled to scratch inside the MFMA loop.
+  const uint32_t wave_u =
+      (uint32_t)__builtin_amdgcn_readfirstlane((int)wave);
+  uint32_t goff_w[2], goff_a[2];
+  #pragma unroll
+  for (int g = 0; g < 2; ++g) {
+    const uint32_t sw = wave_u * 2048 + (uint32_t)g * 1024;
+    const uint32_t sa = (uint32_t)g * 8192 + wave_u * 1024;
+    const uint32_t ow = swz(sw + (uint32_t)lane * 16);
+    const uint32_t oa = swz(sa + (uint32_t)lane * 16);
+    goff_w[g] = (ow >> 7) * (uint32_t)K + ((ow & 127) >> 1);
+    goff_a[g] = (oa >> 7) * (uint32_t)K + ((oa & 127) >> 1);
+  }
+  const uint32_t krow = (uint32_t)K * 128;  // +16384 LDS bytes in elements
+  const uint16_t* Abase = A + (long long)bm * K;
+  const uint16_t* Wbase = W + (long long)bn * K;
+
+  auto glds = [&](const uint16_t* g, uint32_t ldso) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)g,
+        (__attribute__((address_space(3))) uint32_t*)(lds + ldso), 16, 0, 0);
+  };
+  // 16 KiB units: W half h -> pieces (h,g) at goff_w[g] + h*krow;
+  // A unit u (rows [64u,+64) u [128+64u,+64)) -> pieces (u,g) at
+  // goff_a[u] + g*krow
+  auto stage_w_half = [&](int buf, long long kbase, int h) {
+    const uint32_t lb = (uint32_t)buf * (2 * BM * BK) + BM * BK +
+                        ((uint32_t)h * 16384 + wave_u * 2048) / 2;
+    glds(Wbase + kbase + goff_w[0] + (uint32_t)h * krow, lb);
+    glds(Wbase + kbase + goff_w[1] + (uint32_t)h * krow, lb + 512);
+  };
+  auto stage_a_unit = [&](int buf, long long kbase, int u) {
+    const uint32_t lb = (uint32_t)buf * (2 * BM * BK) +
+                        ((uint32_t)u * 8192 + wave_u * 1024) / 2;
+    glds(Abase + kbase + goff_a[u], lb);
+    glds(Abase + kbase + goff_a[u] + krow, lb + 8192);
+  };
+
+  // prologue: tile 0 complete + tile 1 {W.h0, W.h1, A01}; tile 1's A23
+  // is staged by P0 of the first iteration like every later pair (an
+  // unconditional P0 keeps the compiler from peeling iteration 0, which
+  // spilled two accumulator quads into the steady-state loop)
+  stage_w_half(0, 0, 0);
+  stage_w_half(0, 0, 1);
+  stage_a_unit(0, 0, 0);
+  stage_a_unit(0, 0, 1);
+  stage_w_half(1, BK, 0);
+  stage_w_half(1, BK, 1);
+  stage_a_unit(1, BK, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  bf16x8 wfrag[NF][2];
+  for (int kt = 0; kt < KT; kt += 2) {  // iteration = tile pair
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {  // 0: tile a (buf0), 1: b (buf1)
+      const int tile = kt + half;
+      const uint16_t* ldsA = LDSA(half);
+      const uint16_t* ldsW = LDSW(half);
+      const long long knext = (long long)(tile + 2) * BK;
+      const bool stg = (tile + 2 < KT);
+      #pragma unroll
+      for (int q = 0; q < 4; ++q) {  // global phase = half*4 + q
+        // --- fragment ds reads
+        bf16x8 afrag[2][2];
+        if (q == 0) {
+          #pragma unroll
+          for (int j = 0; j < NF; ++j)
+            #pragma unroll
+            for (int kk = 0; kk < 2; ++kk) {
+              const int col = wn * 64 + j * 16 + (lane & 15);
+              const uint32_t off = swz((uint32_t)col * 128 + kk * 64
+                                       + (lane >> 4) * 16);
+              wfrag[j][kk] = *(const bf16x8*)((const char*)ldsW + off);
+            }
+        }
+        #pragma unroll
+        for (int a = 0; a < 2; ++a)
+          #pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            const int row = wm * 128 + (2 * q + a) * 16 + (lane & 15);
+            const uint32_t off = swz((uint32_t)row * 128 + kk * 64
+                                     + (lane >> 4) * 16);
+            afrag[a][kk] = *(const bf16x8*)((const char*)ldsA + off);
+          }
+        // --- the phase's one staging unit
+        if (half == 0) {
+          if (q == 0) {
+            // P0: buf1.A23 for THIS iteration's tile b (unconditional —
+            // the prologue leaves exactly this unit to P0 on pair 0)
+            stage_a_unit(1, (long long)(kt + 1) * BK, 1);
+          } else if (stg) {
+            // P1/P2: W halves of a+2; P3: A01 of a+2
+            if (q <= 2) stage_w_half(0, knext, q - 1);
+            else        stage_a_unit(0, knext, 0);
+          }
+        } else {
+          if (q == 0) {
+            // P4: A23 of a+2 into buf0
+            if (kt + 2 < KT) stage_a_unit(0, (long long)(kt + 2) * BK, 1);
+          } else if (stg) {
+            // P5/P6: W halves of b+2; P7: A01 of b+2
+            if (q <= 2) stage_w_half(1, knext, q - 1);
+            else        stage_a_unit(1, knext, 0);
+          }
+        }
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int a = 0; a < 2; ++a)
+          #pragma unroll
+          for (int j = 0; j < NF; ++j)
+            #pragma unroll
+            for (int kk = 0; kk < 2; ++kk)
+              acc[2 * q + a][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  afrag[a][kk], wfrag[j][kk], acc[2 * q + a][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        if (q == 3) {
+          // end of P3: tile b's units must be landed; end of P7: tile
+          // (a+2)'s units must be landed
+          if (half == 0 ? stg : (kt + 2 < KT))
+            asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+          else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+  }
+
+  #pragma unroll
+  for (int i = 0; i < MF; ++i) {
+    #pragma unroll
+    for (int j = 0; j < NF; ++j) {
+      const int col = bn + wn * 64 + j * 16 + (lane & 15);
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = bm + wm * 128 + i * 16 + 4 * (lane >> 4) + r;
+        C[(long long)row * N + col] = gf2bf(acc[i][j][r]);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------- host
 
 #define HIP_CHECK(x) do { hipError_t e = (x); if (e) { \
@@ -582,23 +785,23 @@ int main() {
     printf("fragment layout assumption failed - skipping GEMM checks\n");
     return 2;
   }
-  // race screen: repeated refchecks at two shapes for both kernels
-  for (int rep = 0; rep < 2; ++rep) {
+  // race screen: repeated refchecks (sync-structure kernels need >=3 reps)
+  for (int rep = 0; rep < 3; ++rep) {
     if (!refcheck(gemm_tn_bf16, "v1", 512, 512, 512)) return 3;
-    if (!refcheck(gemm_tn_bf16, "v1", 1024, 512, 2048)) return 3;
-    if (!refcheck(gemm_tn_bf16_v2, "v2", 512, 512, 512)) return 4;
-    if (!refcheck(gemm_tn_bf16_v2, "v2", 1024, 512, 2048)) return 4;
-    if (!refcheck(gemm_tn_bf16_v2, "v2", 1024, 1024, 4096)) return 4;
-    if (!refcheck(gemm_tn_bf16_v3, "v3", 512, 512, 512)) return 5;
-    if (!refcheck(gemm_tn_bf16_v3, "v3", 1024, 512, 2048)) return 5;
-    if (!refcheck(gemm_tn_bf16_v3, "v3", 1024, 1024, 4096)) return 5;
+    if (!refcheck(gemm_tn_bf16_v4, "v4", 512, 512, 512)) return 6;
+    if (!refcheck(gemm_tn_bf16_v4, "v4", 1024, 512, 2048)) return 6;
+    if (!refcheck(gemm_tn_bf16_v4, "v4", 1024, 1024, 4096)) return 6;
+    if (!refcheck(gemm_tn_bf16_v4, "v4", 2048, 512, 512)) return 6;
   }
   perf(gemm_tn_bf16, "v1", 4096, 4096, 4096, 20);
-  perf(gemm_tn_bf16_v2, "v2", 4096, 4096, 4096, 20);
-  perf(gemm_tn_bf16_v3, "v3", 4096, 4096, 4096, 20);
-  perf(gemm_tn_bf16, "v1", 16384, 8192, 2048, 10);
-  perf(gemm_tn_bf16_v3, "v3", 16384, 8192, 2048, 10);
-  perf(gemm_tn_bf16, "v1", 16384, 2048, 8192, 10);
-  perf(gemm_tn_bf16_v3, "v3", 16384, 2048, 8192, 10);
+  perf(gemm_tn_bf16_v4, "v4", 4096, 4096, 4096, 20);
+  // Llama-8B production shapes (M = 16384 tokens)
+  perf(gemm_tn_bf16, "v1", 16384, 4096, 4096, 10);
+  perf(gemm_tn_bf16_v4, "v4", 16384, 4096, 4096, 10);
+  perf(gemm_tn_bf16, "v1", 16384, 14336, 4096, 10);
+  perf(gemm_tn_bf16_v4, "v4", 16384, 14336, 4096, 10);
+  perf(gemm_tn_bf16, "v1", 16384, 4096, 14336, 10);
+  perf(gemm_tn_bf16_v4, "v4", 16384, 4096, 14336, 10);
+  perf(gemm_tn_bf16_v4, "v4", 8192, 8192, 8192, 10);
   return 0;
 }
