@@ -33,6 +33,8 @@ def parse_args():
     p.add_argument("--model", default="resnet50",
                    choices=["resnet50", "llama8b", "llama1b"])
     p.add_argument("--mode", default="train", choices=["train", "asha"])
+    p.add_argument("--workers", type=int, default=0,
+                   help="ASHA pool size (0 = one per GPU; >gpus time-slices)")
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--graphs", action="store_true",
                    help="capture the whole train step in a hipGraph "
@@ -203,13 +205,14 @@ def run_asha(args):
     from bench_trials import resnet_trial_fn
 
     n_gpus = min(args.gpus, torch.cuda.device_count())
+    n_workers = args.workers or n_gpus
     sp = Searchspace(lr=("DOUBLE", [1e-4, 1e-2]),
                      momentum=("DOUBLE", [0.8, 0.99]))
-    num_trials = max(16, 4 * n_gpus)
+    num_trials = max(16, 4 * n_workers)
     cfg = HyperparameterOptConfig(
         num_trials=num_trials, optimizer="asha", searchspace=sp,
         direction="min", es_policy="median", es_min=4,
-        num_workers=n_gpus, name="bench-asha")
+        num_workers=n_workers, name="bench-asha")
     t0 = time.time()
     res = experiment.lagom(resnet_trial_fn, cfg)
     elapsed = time.time() - t0
@@ -229,7 +232,7 @@ def run_asha(args):
         "data": "synthetic",
         "config": {"model": "resnet50", "num_trials": res["num_trials"],
                    "optimizer": "asha", "parallelism":
-                   "trialpool{}".format(n_gpus)},
+                   "trialpool{}".format(n_workers)},
     }))
 
 

@@ -35,6 +35,10 @@ def _pin_gpu(gpu_id):
     os.environ["HIP_VISIBLE_DEVICES"] = str(gpu_id)
     os.environ["CUDA_VISIBLE_DEVICES"] = str(gpu_id)
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    # deterministic conv-algorithm selection: FAST (immediate mode) skips
+    # the per-shape MIOpen find sweep whose first-trial cost made ASHA
+    # trials/hr vary 2x run to run (round-1 VERDICT weak #5)
+    os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 
 def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload):
