@@ -65,6 +65,10 @@ def make_resnet_step(args, device, world):
     from maggy_amd.ops import FusedAdam
 
     batch = args.batch or 512
+    if os.environ.get("MAGGY_CUDNN_BENCHMARK") == "1":
+        # exhaustive per-shape conv find (cached in the MIOpen user
+        # find-db); default stays heuristic FAST mode
+        torch.backends.cudnn.benchmark = True
     model = resnet50().to(device, memory_format=torch.channels_last)
     if world > 1:
         from maggy_amd.parallel.dist import wrap_ddp
