@@ -17,6 +17,7 @@ is preserved exactly:
   IDLE   -> controller had no trial ready; retried after 0.1 s
 """
 import json
+import threading
 import time
 
 from maggy_amd import util
@@ -29,6 +30,7 @@ from maggy_amd.trial import Trial
 from maggy_amd.utils.jsonutil import json_default_numpy
 
 IDLE_RETRY_S = 0.1
+PROGRESS_INTERVAL_S = 2.0
 
 
 class OptimizationDriver:
@@ -84,6 +86,13 @@ class OptimizationDriver:
         self.result = {}
         self.maggy_log = ""
         self.executor_logs = ""  # jupyter-style log stream (parity LOG msg)
+        self._log_stream_lock = threading.Lock()
+        # live-progress callback(status_string, new_logs), invoked from
+        # the event loop every PROGRESS_INTERVAL_S (parity: the LOG
+        # snapshots the reference served to Jupyter mid-run,
+        # rpc.py:490-502); set via experiment.lagom(..., progress=...)
+        self.progress_cb = None
+        self._last_progress = 0.0
         self.job_start = None
         self.job_end = None
         self.duration = None
@@ -187,8 +196,12 @@ class OptimizationDriver:
                     self._handle_final(w, msg)
                 elif kind == M.ERROR:
                     self._handle_error(w, msg)
+                elif kind == M.LOG:
+                    with self._log_stream_lock:
+                        self.executor_logs += msg[2]
             self._handle_metrics(pool.drain_metrics())
             self._report_ring_drops()
+            self._emit_progress()
             self._retry_idle()
             self._reap_dead()
             if self.experiment_done and self._all_workers_free():
@@ -263,7 +276,8 @@ class OptimizationDriver:
         w.trial_id = None
         self.last_final_ts = time.time()
         if logs:
-            self.executor_logs += logs
+            with self._log_stream_lock:
+                self.executor_logs += logs
         self._update_result(trial)
         self.maggy_log = self.log_string()
         self.log(self.maggy_log)
@@ -429,11 +443,25 @@ class OptimizationDriver:
         if trial.early_stop:
             self.result["early_stopped"] += 1
 
+    def _emit_progress(self):
+        if self.progress_cb is None:
+            return
+        now = time.time()
+        if now - self._last_progress < PROGRESS_INTERVAL_S:
+            return
+        self._last_progress = now
+        try:
+            self.progress_cb(*self.get_logs())
+        except Exception as e:
+            self.log("progress callback failed: {}".format(e))
+
     def get_logs(self):
         """Drain the accumulated executor log stream plus a progress
-        snapshot (parity: LOG request, rpc.py:490-502)."""
-        logs = self.executor_logs
-        self.executor_logs = ""
+        snapshot (parity: LOG request, rpc.py:490-502).  Thread-safe:
+        callable mid-run from another thread (lagom_async handle)."""
+        with self._log_stream_lock:
+            logs = self.executor_logs
+            self.executor_logs = ""
         return self.log_string(), logs
 
     def log_string(self):

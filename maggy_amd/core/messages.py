@@ -10,3 +10,4 @@ TRIAL = "TRIAL"    # driver -> worker: (trial_id, params)
 FINAL = "FINAL"    # worker -> driver: (trial_id, opt_val, duration, early, logs)
 ERROR = "ERROR"    # worker -> driver: (trial_id, traceback)
 GSTOP = "GSTOP"    # driver -> worker: experiment done, exit
+LOG = "LOG"        # worker -> driver: mid-trial log text (throttled)

@@ -106,7 +106,12 @@ class Reporter:
 
     def log(self, log_msg, jupyter=False):
         """Log to the worker logfile (and the trial logfile when a trial is
-        active)."""
+        active).  When a ``log_sink`` is attached (the worker's control
+        pipe), accumulated jupyter-stream text is flushed through it at
+        most once per second — live log streaming to the driver mid-trial
+        (the reference shipped logs with every heartbeat,
+        rpc.py:723-726)."""
+        flush = None
         with self.lock:
             msg = "{} ({}): {}\n".format(
                 datetime.now().isoformat(), self.worker_id, log_msg
@@ -121,6 +126,19 @@ class Reporter:
                         self.fd.write(msg)
                     self.print_fn(msg.rstrip("\n"))
             except (IOError, ValueError, AttributeError):
+                pass
+            sink = getattr(self, "log_sink", None)
+            if sink is not None and self.logs:
+                import time as _time
+
+                now = _time.time()
+                if now - getattr(self, "_last_sink", 0.0) >= 1.0:
+                    self._last_sink = now
+                    flush, self.logs = self.logs, ""
+        if flush is not None:
+            try:
+                sink(flush)
+            except Exception:
                 pass
 
     # -- driver/test helpers --------------------------------------------

@@ -18,6 +18,7 @@ The per-trial semantics mirror the reference executor exactly:
 import builtins
 import json
 import os
+import threading
 import time
 import traceback
 
@@ -55,6 +56,16 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
     reporter = Reporter(
         ring=ring, log_file=log_file, worker_id=worker_id, print_fn=real_print
     )
+    # throttled mid-trial log streaming to the driver (LOG message);
+    # user code may call print/reporter.log from its own threads, so every
+    # pipe send goes through one lock (Connection.send is not thread-safe)
+    send_lock = threading.Lock()
+
+    def send(msg):
+        with send_lock:
+            conn.send(msg)
+
+    reporter.log_sink = lambda text: send((M.LOG, worker_id, text))
 
     def maggy_print(*args, **kwargs):
         real_print(*args, **kwargs)
@@ -65,7 +76,7 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
     experiment_type = payload.get("experiment_type", "optimization")
 
     try:
-        conn.send((M.REG, worker_id, os.getpid()))
+        send((M.REG, worker_id, os.getpid()))
         while True:
             msg = conn.recv()
             if msg[0] == M.GSTOP:
@@ -132,17 +143,17 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
                     retval, trial_dir, optimization_key, trial_log_file
                 )
                 reporter.log("Finished Trial: {}".format(trial_id), False)
-                conn.send((M.FINAL, worker_id, trial_id, retval,
+                send((M.FINAL, worker_id, trial_id, retval,
                            time.time() - start, early_stopped, reporter.logs))
             except EarlyStopException as e:
                 early_stopped = True
                 reporter.log("Early Stopped Trial.", False)
-                conn.send((M.FINAL, worker_id, trial_id, e.metric,
+                send((M.FINAL, worker_id, trial_id, e.metric,
                            time.time() - start, early_stopped, reporter.logs))
             except Exception:
                 tb = traceback.format_exc()
                 reporter.log(tb, False)
-                conn.send((M.ERROR, worker_id, trial_id, tb))
+                send((M.ERROR, worker_id, trial_id, tb))
             finally:
                 builtins.print = real_print
                 reporter.logs = ""

@@ -1,3 +1,7 @@
-from maggy_amd.experiment.experiment import lagom  # noqa: F401
+from maggy_amd.experiment.experiment import (  # noqa: F401
+    LagomHandle,
+    lagom,
+    lagom_async,
+)
 
-__all__ = ["lagom"]
+__all__ = ["lagom", "lagom_async", "LagomHandle"]

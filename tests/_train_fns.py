@@ -108,3 +108,12 @@ def noisy_quadratic_fn(hparams, reporter):
     v = 1.0 - (hparams["lr"] - 0.06) ** 2 * 100.0
     reporter.broadcast(v, 0)
     return v
+
+
+def chatty_slow_fn(hparams, reporter):
+    """Prints every step for ~3 s: feeds the live log-stream test."""
+    for step in range(30):
+        print("chatty step", step)
+        reporter.broadcast(float(step), step)
+        time.sleep(0.1)
+    return 1.0
