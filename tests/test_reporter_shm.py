@@ -138,10 +138,14 @@ def test_ring_threaded_producer_consumer_stress():
         t = threading.Thread(target=produce)
         t.start()
         seen = []
-        while t.is_alive() or True:
+        while True:
+            # read liveness BEFORE draining: if the producer was already
+            # dead, this drain observes every push (GIL ordering); checking
+            # after would race a final burst of pushes
+            alive = t.is_alive()
             recs = ring.drain()
             seen.extend(recs)
-            if not t.is_alive() and not recs:
+            if not alive and not recs:
                 break
         steps = [r[1] for r in seen]
         # strictly increasing (no duplication, no reordering)
