@@ -79,14 +79,25 @@ class MetricRing:
         if head == self._tail:
             return []
         start = max(self._tail, head - self.slots)
-        if start > self._tail:
-            self.dropped += start - self._tail
         if max_records is not None:
             start = max(start, head - max_records)
         out = []
         for i in range(start, head):
             off = HEADER_BYTES + (i % self.slots) * RECORD_BYTES
             out.append(struct.unpack_from(RECORD_FMT, self.shm.buf, off))
+        # seqlock-style re-validation: a record read above may have been
+        # OVERWRITTEN mid-drain by a producer that lapped the ring (torn
+        # read).  Slot i is overwritten while writing record i+slots,
+        # which can be in progress before head advances past it — so
+        # after re-reading head, only records with i > head2 - slots are
+        # guaranteed stable; earlier ones are discarded as dropped.
+        head2 = struct.unpack_from("<q", self.shm.buf, 0)[0]
+        cut = head2 - self.slots
+        if cut >= start:
+            n_cut = min(cut + 1, head) - start
+            del out[:n_cut]
+            start += n_cut
+        self.dropped += start - self._tail
         self._tail = head
         return out
 
