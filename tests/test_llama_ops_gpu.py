@@ -100,3 +100,41 @@ def test_tiny_llama_gpu_trains():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0] - 0.5, losses
+
+
+@pytest.mark.gpu
+def test_sdpa_efficient_backward_matches_math():
+    """The CK memory-efficient attention backward (the training default,
+    +14% tokens/sec over flash) must match the math backend's gradients."""
+    import os
+
+    from maggy_amd.models.llama import _sdpa
+
+    torch.manual_seed(0)
+    B, H, KV, T, D = 2, 8, 2, 128, 64
+    q = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, KV, T, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, KV, T, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    dy = torch.randn(B, H, T, D, device="cuda", dtype=torch.bfloat16)
+
+    out = _sdpa(q, k, v, causal=True, gqa=True)  # auto-selects efficient
+    out.backward(dy)
+    grads = [t.grad.float().clone() for t in (q, k, v)]
+    for t in (q, k, v):
+        t.grad = None
+
+    os.environ["MAGGY_SDPA"] = "math"
+    try:
+        out2 = _sdpa(q, k, v, causal=True, gqa=True)
+        out2.backward(dy)
+    finally:
+        del os.environ["MAGGY_SDPA"]
+    for got, t in zip(grads, (q, k, v)):
+        ref = t.grad.float()
+        err = (got - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1e-6
+        assert err / scale < 0.06, err / scale
+    assert (out.float() - out2.float()).abs().max().item() < 0.1
