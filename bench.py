@@ -99,7 +99,7 @@ def make_llama_step(args, device, world, size):
 
     cfg_model = (LlamaConfig.llama3_8b() if size == "8b"
                  else LlamaConfig.small_1b())
-    batch = args.batch or (4 if size == "8b" else 8)
+    batch = args.batch or (6 if size == "8b" else 8)
     seq = args.seq_len
     # construct directly on the GPU (init kernels run on-device; 8B fp32
     # transient fits easily in 288 GB HBM3E), then cast params to bf16
@@ -245,6 +245,11 @@ if __name__ == "__main__":
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
     args = parse_args()
+    if args.model.startswith("llama") and not args.graphs:
+        # llama-8B at b6 sits near the 288 GB ceiling; expandable
+        # segments avoid fragmentation OOM (not graph-compatible)
+        os.environ.setdefault("PYTORCH_ALLOC_CONF",
+                              "expandable_segments:True")
     if args.mode == "asha":
         run_asha(args)
     else:

@@ -15,7 +15,11 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from maggy_amd.ops.fused_rms import MaggyRMSNorm, swiglu  # noqa: F401
+from maggy_amd.ops.fused_rms import (  # noqa: F401
+    MaggyRMSNorm,
+    rope_bthd,
+    swiglu,
+)
 from maggy_amd.ops.linear import MaggyFeedForward, MaggyLinear
 
 
@@ -125,11 +129,13 @@ class Attention(nn.Module):
 
     def forward(self, x, cos, sin, cache=None, pos=0):
         B, T, _ = x.shape
-        q = self.wq(x).view(B, T, self.n_heads, self.head_dim).transpose(1, 2)
-        k = self.wk(x).view(B, T, self.n_kv_heads, self.head_dim).transpose(1, 2)
+        # RoPE applies on the [B, T, H, D] layout BEFORE the head
+        # transpose (one fused bf16 pass, ops/fused_rms.py::rope_bthd)
+        q = self.wq(x).view(B, T, self.n_heads, self.head_dim)
+        k = self.wk(x).view(B, T, self.n_kv_heads, self.head_dim)
         v = self.wv(x).view(B, T, self.n_kv_heads, self.head_dim).transpose(1, 2)
-        q = apply_rope(q, cos, sin, pos)
-        k = apply_rope(k, cos, sin, pos)
+        q = rope_bthd(q, cos, sin, pos).transpose(1, 2)
+        k = rope_bthd(k, cos, sin, pos).transpose(1, 2)
         if cache is not None:
             # cache: dict with "k"/"v" [B, KV, T_past, D] (decode path)
             if cache.get("k") is not None:

@@ -47,6 +47,9 @@ extern "C" void launch_gemm_tn_bf16_swiglu(const void*, const void*, void*,
                                            int, hipStream_t);
 extern "C" void launch_transpose_bf16(const void*, void*, int, int,
                                       hipStream_t);
+extern "C" void launch_rope(const void*, void*, const void*, const void*,
+                            long long, int, int, int, int, float,
+                            hipStream_t);
 
 namespace {
 
@@ -265,6 +268,23 @@ torch::Tensor transpose2d(torch::Tensor in) {
   return out;
 }
 
+// RoPE on [B, T, H, D] bf16 (t = dim 1), fp32 cos/sin [>=pos+T, D/2];
+// sign=+1 forward rotation, -1 backward
+torch::Tensor rope(torch::Tensor x, torch::Tensor cost, torch::Tensor sint,
+                   int64_t pos, double sign) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 &&
+              x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(cost.dtype() == torch::kFloat32 && cost.is_contiguous());
+  const int64_t B = x.size(0), T = x.size(1), H = x.size(2), D = x.size(3);
+  TORCH_CHECK(D % 4 == 0, "rope needs head_dim % 4 == 0");
+  TORCH_CHECK(cost.size(-1) == D / 2, "cos table width must be D/2");
+  auto out = torch::empty_like(x);
+  launch_rope(x.data_ptr(), out.data_ptr(), cost.data_ptr(),
+              sint.data_ptr(), B * T * H, (int)D, (int)H, (int)T, (int)pos,
+              (float)sign, current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -289,4 +309,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn_swiglu", &gemm_tn_swiglu,
         "bf16 MFMA GEMM y3 = A @ W3^T with fused h = silu(y1)*y3");
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
+  m.def("rope", &rope, "fused rotary embedding (bf16, fp32 tables)");
 }

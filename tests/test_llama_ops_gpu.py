@@ -138,3 +138,33 @@ def test_sdpa_efficient_backward_matches_math():
         scale = ref.abs().max().item() + 1e-6
         assert err / scale < 0.06, err / scale
     assert (out.float() - out2.float()).abs().max().item() < 0.1
+
+
+@pytest.mark.gpu
+def test_rope_kernel_matches_eager():
+    """Fused RoPE (fwd sign=+1, bwd sign=-1) vs the eager fp32 rotation."""
+    from maggy_amd.models.llama import precompute_rope
+    from maggy_amd.ops.fused_rms import rope_bthd
+
+    torch.manual_seed(0)
+    B, T, H, D = 2, 64, 4, 32
+    cos, sin = precompute_rope(D, 128, 10000.0)
+    cos, sin = cos.cuda(), sin.cuda()
+    for pos in (0, 17):
+        x = torch.randn(B, T, H, D, device="cuda",
+                        dtype=torch.bfloat16).requires_grad_(True)
+        out = rope_bthd(x, cos, sin, pos)
+        dy = torch.randn_like(out)
+        out.backward(dy)
+
+        xr = x.detach().float().clone().requires_grad_(True)
+        c = cos[pos:pos + T][None, :, None, :]
+        s = sin[pos:pos + T][None, :, None, :]
+        x1, x2 = xr[..., 0::2], xr[..., 1::2]
+        ref = torch.empty_like(xr)
+        ref[..., 0::2] = x1 * c - x2 * s
+        ref[..., 1::2] = x2 * c + x1 * s
+        ref.backward(dy.float())
+
+        assert (out.float() - ref.detach()).abs().max().item() < 0.03
+        assert (x.grad.float() - xr.grad).abs().max().item() < 0.03
