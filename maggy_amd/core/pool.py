@@ -65,14 +65,22 @@ class TrialPool:
             self._spawn(w)
 
     def _spawn(self, w):
+        import os
+
         w.ring = MetricRing(slots=self.ring_slots, create=True)
         parent_conn, child_conn = self.ctx.Pipe()
         w.conn = parent_conn
         w.registered = False
+        # forkserver children inherit the fork SERVER's environment
+        # (frozen at server start), not the driver's current one — ship a
+        # live snapshot so user code reading os.environ sees spawn
+        # semantics
+        payload = dict(self.payload)
+        payload["_env"] = dict(os.environ)
         w.process = self.ctx.Process(
             target=worker_main,
             args=(w.worker_id, w.gpu_id, child_conn, w.ring.name,
-                  self.ring_slots, self.log_dir, self.payload),
+                  self.ring_slots, self.log_dir, payload),
             daemon=True,
         )
         w.process.start()

@@ -49,6 +49,9 @@ def worker_main(worker_id, gpu_id, conn, ring_name, ring_slots, log_dir, payload
     {train_fn, model, dataset, optimization_key, experiment_type,
      dataset_generator, model_generator}.
     """
+    env = payload.pop("_env", None)
+    if env:
+        os.environ.update(env)  # forkserver: restore the driver's env
     _pin_gpu(gpu_id)
     ring = MetricRing(name=ring_name, slots=ring_slots)
     log_file = os.path.join(log_dir, "executor_{}.log".format(worker_id))
