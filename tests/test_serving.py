@@ -1,6 +1,8 @@
 """Dynamic-batching model server tests (CPU)."""
 import threading
 
+import pytest
+
 import torch
 
 from maggy_amd.models import MLP
@@ -82,3 +84,55 @@ def test_generation_serving():
     for p, o in zip(prompts, outs):
         assert o.shape == (10,)
         assert torch.equal(o[:6], p)
+
+
+def _replica_model_fn():
+    """Builds the 'model' inside the replica (picklable, zero-arg)."""
+    import torch
+
+    def predict(batch):
+        return batch * 2.0 + 1.0
+
+    return predict
+
+
+@pytest.mark.timeout(120)
+def test_replicated_server_routes_and_serves():
+    from maggy_amd.serving import ReplicatedModelServer
+
+    with ReplicatedModelServer(_replica_model_fn, n_replicas=2,
+                               gpu_ids=[None, None], max_batch=8,
+                               max_wait_ms=10.0) as srv:
+        futs = [srv.submit(torch.full((3,), float(i))) for i in range(24)]
+        for i, f in enumerate(futs):
+            out = f.result(timeout=30)
+            assert torch.allclose(out, torch.full((3,), float(i) * 2 + 1))
+        assert srv.stats["requests"] == 24
+        # least-loaded routing actually used both replicas
+        assert all(n > 0 for n in srv.stats["per_replica"]), \
+            srv.stats["per_replica"]
+
+
+@pytest.mark.timeout(120)
+def test_replicated_server_error_propagates():
+    from maggy_amd.serving import ReplicatedModelServer
+
+    with ReplicatedModelServer(_failing_model_fn, n_replicas=1,
+                               gpu_ids=[None], max_batch=4,
+                               max_wait_ms=5.0) as srv:
+        with pytest.raises(RuntimeError, match="boom"):
+            srv.predict(torch.ones(2), timeout=30)
+        # the replica survives a failing batch
+        out = srv.predict(torch.zeros(2), timeout=30)
+        assert torch.allclose(out, torch.zeros(2))
+
+
+def _failing_model_fn():
+    import torch
+
+    def predict(batch):
+        if batch.sum() > 0:
+            raise ValueError("boom")
+        return batch
+
+    return predict
