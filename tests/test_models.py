@@ -70,3 +70,21 @@ def test_small_transformer():
     y = m(tokens)
     assert y.shape == (4, 3)
     y.sum().backward()
+
+
+def test_rope_bthd_matches_legacy_apply_rope():
+    """rope_bthd on [B,T,H,D] must equal the original apply_rope on
+    [B,H,T,D] (the layouts commute with the transpose)."""
+    import torch
+
+    from maggy_amd.models.llama import apply_rope, precompute_rope
+    from maggy_amd.ops.fused_rms import rope_bthd
+
+    torch.manual_seed(0)
+    B, T, H, D = 2, 12, 3, 16
+    cos, sin = precompute_rope(D, 32, 10000.0)
+    x = torch.randn(B, T, H, D)
+    for pos in (0, 5):
+        got = rope_bthd(x, cos, sin, pos).transpose(1, 2)
+        ref = apply_rope(x.transpose(1, 2), cos, sin, pos)
+        assert torch.allclose(got, ref, atol=1e-6)

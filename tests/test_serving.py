@@ -136,3 +136,32 @@ def _failing_model_fn():
         return batch
 
     return predict
+
+
+def _gpu_model_fn():
+    import torch
+
+    from maggy_amd.models import MLP
+
+    torch.manual_seed(0)
+    model = MLP(in_features=16, hidden=32, num_classes=4).cuda().eval()
+
+    def predict(batch):
+        return model(batch)
+
+    return predict
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(180)
+def test_replicated_server_gpu_replica():
+    """One CUDA replica process serving a real model."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from maggy_amd.serving import ReplicatedModelServer
+
+    with ReplicatedModelServer(_gpu_model_fn, n_replicas=1, gpu_ids=[0],
+                               max_batch=16, max_wait_ms=5.0) as srv:
+        outs = [srv.predict(torch.randn(16), timeout=60) for _ in range(8)]
+        assert all(o.shape == (4,) for o in outs)
+        assert srv.stats["per_replica"][0] == 8
