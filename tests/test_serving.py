@@ -165,3 +165,33 @@ def test_replicated_server_gpu_replica():
         outs = [srv.predict(torch.randn(16), timeout=60) for _ in range(8)]
         assert all(o.shape == (4,) for o in outs)
         assert srv.stats["per_replica"][0] == 8
+
+
+def _dying_model_fn():
+    import os
+
+    import torch  # noqa: F401
+
+    def predict(batch):
+        os._exit(17)  # hard replica death mid-batch
+
+    return predict
+
+
+@pytest.mark.timeout(120)
+def test_replicated_server_replica_death_times_out_cleanly():
+    """A replica that dies hard must not hang the client: the future
+    times out and the server still shuts down."""
+    from concurrent.futures import TimeoutError as FutTimeout
+
+    from maggy_amd.serving import ReplicatedModelServer
+
+    srv = ReplicatedModelServer(_dying_model_fn, n_replicas=1,
+                                gpu_ids=[None], max_batch=4,
+                                max_wait_ms=5.0).start()
+    try:
+        fut = srv.submit(torch.ones(2))
+        with pytest.raises(FutTimeout):
+            fut.result(timeout=3)
+    finally:
+        srv.stop()
