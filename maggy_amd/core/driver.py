@@ -208,9 +208,14 @@ class OptimizationDriver:
                 break
             if not any(w.process is not None and w.process.is_alive()
                        for w in pool.workers):
-                from maggy_amd.exceptions import WorkerCrashError
+                # a worker that died AFTER this tick's _reap_dead() is
+                # respawned on the next tick — give up only when no
+                # worker is respawnable any more (TOCTOU flake otherwise)
+                if all(w.process is None or w.respawns >= 3
+                       for w in pool.workers):
+                    from maggy_amd.exceptions import WorkerCrashError
 
-                raise WorkerCrashError("all", "all trial workers died")
+                    raise WorkerCrashError("all", "all trial workers died")
 
     def _all_workers_free(self):
         return all(w.trial_id is None for w in self.pool.workers)
