@@ -94,6 +94,16 @@ def main():
         for _ in range(it):
             ext.gemm_tn(a, w)
 
+    if on("rope"):
+        from maggy_amd.ops.fused_rms import rope_bthd
+        from maggy_amd.models.llama import precompute_rope
+
+        cos, sin = precompute_rope(128, 4096, 500000.0)
+        cos, sin = cos.cuda(), sin.cuda()
+        x = torch.randn(4, 4096, 32, 128, device=dev).bfloat16()
+        for _ in range(it):
+            rope_bthd(x, cos, sin, 0)
+
     if on("transpose"):
         x = torch.randn(16384, 4096, device=dev).bfloat16()
         for _ in range(it):
