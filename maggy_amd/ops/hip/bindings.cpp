@@ -275,9 +275,13 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cost, torch::Tensor sint,
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 &&
               x.is_contiguous() && x.dim() == 4);
   TORCH_CHECK(cost.dtype() == torch::kFloat32 && cost.is_contiguous());
+  TORCH_CHECK(sint.dtype() == torch::kFloat32 && sint.is_contiguous());
   const int64_t B = x.size(0), T = x.size(1), H = x.size(2), D = x.size(3);
   TORCH_CHECK(D % 4 == 0, "rope needs head_dim % 4 == 0");
-  TORCH_CHECK(cost.size(-1) == D / 2, "cos table width must be D/2");
+  TORCH_CHECK(cost.size(-1) == D / 2 && sint.size(-1) == D / 2,
+              "cos/sin table width must be D/2");
+  TORCH_CHECK(cost.size(0) >= pos + T && sint.size(0) >= pos + T,
+              "rope tables too short: need pos+T rows");
   auto out = torch::empty_like(x);
   launch_rope(x.data_ptr(), out.data_ptr(), cost.data_ptr(),
               sint.data_ptr(), B * T * H, (int)D, (int)H, (int)T, (int)pos,
