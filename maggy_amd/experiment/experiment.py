@@ -85,18 +85,15 @@ def lagom_async(train_fn, config=None):
     import threading
 
     box = {}
-
-    def _capture_driver(status, logs):
-        # progress callback doubles as the driver-handle hook; the box is
-        # filled by _lagom_driver below before the loop starts
-        return
-
     handle_holder = {}
 
     def _run():
         h = handle_holder["h"]
         try:
-            h._result = lagom(train_fn, config, progress=_capture_driver)
+            # no progress callback: _emit_progress would DRAIN the log
+            # stream into it, racing the handle's own get_logs(); the
+            # driver is published to the box by _lagom_driver regardless
+            h._result = lagom(train_fn, config)
         except BaseException as e:  # surfaced via result()
             h._exc = e
 
