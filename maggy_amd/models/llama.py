@@ -137,11 +137,22 @@ class Attention(nn.Module):
         q = rope_bthd(q, cos, sin, pos).transpose(1, 2)
         k = rope_bthd(k, cos, sin, pos).transpose(1, 2)
         if cache is not None:
-            # cache: dict with "k"/"v" [B, KV, T_past, D] (decode path)
-            if cache.get("k") is not None:
-                k = torch.cat([cache["k"], k], dim=2)
-                v = torch.cat([cache["v"], v], dim=2)
-            cache["k"], cache["v"] = k, v
+            # STATIC KV cache (decode path): preallocated [B, KV, cap, D]
+            # buffers written in place — the torch.cat idiom re-copies the
+            # whole cache every decoded token (~1 GB/token at 8B b8)
+            if cache.get("kbuf") is None:
+                cap = int(cache.get("cap", pos + T + 256))
+                cache["kbuf"] = k.new_empty(
+                    B, self.n_kv_heads, cap, self.head_dim)
+                cache["vbuf"] = v.new_empty(
+                    B, self.n_kv_heads, cap, self.head_dim)
+                cache["len"] = 0
+            L = cache["len"]
+            cache["kbuf"][:, :, L:L + T] = k
+            cache["vbuf"][:, :, L:L + T] = v
+            cache["len"] = L + T
+            k = cache["kbuf"][:, :, :L + T]
+            v = cache["vbuf"][:, :, :L + T]
         # causal masking is needed only when the query block spans >1 new
         # position; a single decoded token attends to the whole cache
         causal = T > 1
@@ -219,7 +230,8 @@ class LlamaModel(nn.Module):
         :returns: [B, T_prompt + max_new_tokens] ids
         """
         self.eval()
-        caches = [{"k": None, "v": None} for _ in self.layers]
+        cap = tokens.shape[1] + max_new_tokens
+        caches = [{"kbuf": None, "cap": cap} for _ in self.layers]
         out = tokens
         x_in = tokens
         pos = 0
