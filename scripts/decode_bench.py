@@ -21,6 +21,8 @@ def main():
     ap.add_argument("--prompt", type=int, default=512)
     ap.add_argument("--new", type=int, default=64)
     ap.add_argument("--iters", type=int, default=3)
+    ap.add_argument("--captured", action="store_true",
+                    help="hipGraph-captured decode step")
     args = ap.parse_args()
     cfg = (LlamaConfig.llama3_8b() if args.size == "8b"
            else LlamaConfig.small_1b())
@@ -31,18 +33,20 @@ def main():
     model.rope_sin = model.rope_sin.float()
     tokens = torch.randint(0, cfg.vocab_size, (args.batch, args.prompt),
                            device="cuda")
+    gen = (lambda t, n: model.generate_captured(t, n)) if args.captured \
+        else (lambda t, n: model.generate(t, n))
     # warmup
-    model.generate(tokens, max_new_tokens=8)
+    gen(tokens, 8)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.iters):
-        model.generate(tokens, max_new_tokens=args.new)
+        gen(tokens, args.new)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.iters
     new_tok = args.batch * args.new
-    print("llama-%s decode: batch %d, prompt %d, %d new tokens: "
+    print("llama-%s decode%s: batch %d, prompt %d, %d new tokens: "
           "%.3f s -> %.0f tokens/sec decode (%.2f ms/token/batch)"
-          % (args.size, args.batch, args.prompt, args.new, dt,
+          % (args.size, " (captured)" if args.captured else "", args.batch, args.prompt, args.new, dt,
              new_tok / dt, dt / args.new * 1e3))
 
 

@@ -47,3 +47,19 @@ def test_training_forward_unchanged():
     tokens = torch.randint(0, 97, (2, 8))
     loss = m(tokens, tokens)
     assert torch.isfinite(loss)
+
+
+def test_captured_path_matches_generate_cpu():
+    """The static-buffer masked decode step (the hipGraph-capturable
+    path, run eagerly on CPU) must produce the same tokens as the
+    regular KV-cached generate."""
+    import torch
+
+    from maggy_amd.models import LlamaConfig, LlamaModel
+
+    torch.manual_seed(0)
+    m = LlamaModel(LlamaConfig.tiny(vocab_size=97)).eval()
+    prompt = torch.randint(0, 97, (3, 9))
+    ref = m.generate(prompt, max_new_tokens=7)
+    got = m.generate_captured(prompt, max_new_tokens=7, use_graph=False)
+    assert torch.equal(got, ref)

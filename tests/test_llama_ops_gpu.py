@@ -168,3 +168,20 @@ def test_rope_kernel_matches_eager():
 
         assert (out.float() - ref.detach()).abs().max().item() < 0.03
         assert (x.grad.float() - xr.grad).abs().max().item() < 0.03
+
+
+@pytest.mark.gpu
+def test_generate_captured_graph_matches_eager():
+    """hipGraph-captured decode == eager KV-cached decode on GPU."""
+    from maggy_amd.models import LlamaConfig, LlamaModel
+
+    torch.manual_seed(0)
+    with torch.device("cuda"):
+        m = LlamaModel(LlamaConfig.tiny(vocab_size=97))
+    m = m.to(torch.bfloat16).eval()
+    m.rope_cos = m.rope_cos.float()
+    m.rope_sin = m.rope_sin.float()
+    prompt = torch.randint(0, 97, (2, 8), device="cuda")
+    ref = m.generate(prompt, max_new_tokens=6)
+    got = m.generate_captured(prompt, max_new_tokens=6, use_graph=True)
+    assert torch.equal(got, ref)
